@@ -1,0 +1,835 @@
+// bifrost_amd: bfLinAlgMatMul — the hot path.
+//
+// Behaviour contract (re-derived from the reference, not translated):
+//   bfLinAlgMatMul(alpha, a, b, beta, c):
+//     a&&b   -> c = alpha * a.b + beta * c                (bfMatMul_ab)
+//     b only -> c = alpha * b^H.b + beta * c  (adjoint)   (bfMatMul_aa)
+//     a only -> c = alpha * a.a^H + beta * c              (bfMatMul_aa)
+//   For the herk forms only the LOWER triangle of c (row-major) is written;
+//   beta==0 means c is not read.  Batch dims flatten and the largest one
+//   becomes the kernel batch dim; stride analysis decides the layout case
+//   (reference src/linalg.cu:242-357,723-862 semantics).
+//
+// Kernels (all our own, CDNA4-first):
+//   cherk_ci8_kernel     — specialized per-channel ci8 X^H.X accumulation,
+//                          k-major layout (the correlator): 256-thread
+//                          blocks, 32x32 complex output tiles (2x2 per
+//                          lane), LDS-staged K slabs, fp32 accumulation,
+//                          triangular block lift.  (v1: VALU fp32 MACs;
+//                          i8-MFMA tiles are the planned v2.)
+//   herk_generic_kernel  — any dtype/layout herk fallback, 16x16 LDS tiles.
+//   gemm_generic_kernel  — any dtype/layout gemm fallback, 16x16 LDS tiles.
+//   beamform_kernel      — B = W.X with W [beam][k] k-fast cached in LDS,
+//                          X [t][k] k-fast streamed, fp32 accum; nbeam
+//                          tiles of <=16 (reference caps nbeam at 16;
+//                          we chunk so any nbeam works, config 5 needs 64).
+
+#include <bifrost/linalg.h>
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cmath>
+#include <cstring>
+
+#include "dtype.hpp"
+#include "hipctx.hpp"
+#include "status.hpp"
+
+namespace {
+
+using bfamd::StatusError;
+
+/* ------------------------------ helpers ------------------------------- */
+
+struct f2 { float x, y; };
+struct d2 { double x, y; };
+
+template <typename T> struct AccOf;
+template <> struct AccOf<float>  { using type = float;  using real = float; static constexpr bool cplx = false; };
+template <> struct AccOf<double> { using type = double; using real = double; static constexpr bool cplx = false; };
+template <> struct AccOf<f2>     { using type = f2;     using real = float; static constexpr bool cplx = true; };
+template <> struct AccOf<d2>     { using type = d2;     using real = double; static constexpr bool cplx = true; };
+
+// Element loaders: convert a storage element to the accumulator type.
+template <typename Acc> struct LoadCI8 {
+    __device__ static Acc load(const void* p, long idx) {
+        const signed char* q = (const signed char*)p + 2 * idx;
+        return Acc{(typename AccOf<Acc>::real)q[0],
+                   (typename AccOf<Acc>::real)q[1]};
+    }
+};
+template <typename Acc> struct LoadCI16 {
+    __device__ static Acc load(const void* p, long idx) {
+        const short* q = (const short*)p + 2 * idx;
+        return Acc{(typename AccOf<Acc>::real)q[0],
+                   (typename AccOf<Acc>::real)q[1]};
+    }
+};
+// ci4: one byte per complex, re in the HIGH nibble (src/Complex.hpp:149-168)
+template <typename Acc> struct LoadCI4 {
+    __device__ static Acc load(const void* p, long idx) {
+        signed char b = ((const signed char*)p)[idx];
+        return Acc{(typename AccOf<Acc>::real)(b >> 4),
+                   (typename AccOf<Acc>::real)((signed char)(b << 4) >> 4)};
+    }
+};
+template <typename Acc> struct LoadCF32 {
+    __device__ static Acc load(const void* p, long idx) {
+        const float* q = (const float*)p + 2 * idx;
+        return Acc{(typename AccOf<Acc>::real)q[0],
+                   (typename AccOf<Acc>::real)q[1]};
+    }
+};
+template <typename Acc> struct LoadCF64 {
+    __device__ static Acc load(const void* p, long idx) {
+        const double* q = (const double*)p + 2 * idx;
+        return Acc{(typename AccOf<Acc>::real)q[0],
+                   (typename AccOf<Acc>::real)q[1]};
+    }
+};
+template <typename Acc> struct LoadF32 {
+    __device__ static Acc load(const void* p, long idx) {
+        return (Acc)((const float*)p)[idx];
+    }
+};
+template <typename Acc> struct LoadF64 {
+    __device__ static Acc load(const void* p, long idx) {
+        return (Acc)((const double*)p)[idx];
+    }
+};
+
+// Complex MAC: acc += conj?(a) * conj?(b)
+template <bool CA, bool CB>
+__device__ inline void cmac(f2& acc, f2 a, f2 b) {
+    float aim = CA ? -a.y : a.y;
+    float bim = CB ? -b.y : b.y;
+    acc.x = fmaf(a.x, b.x, acc.x);
+    acc.x = fmaf(-aim, bim, acc.x);
+    acc.y = fmaf(a.x, bim, acc.y);
+    acc.y = fmaf(aim, b.x, acc.y);
+}
+template <bool CA, bool CB>
+__device__ inline void cmac(d2& acc, d2 a, d2 b) {
+    double aim = CA ? -a.y : a.y;
+    double bim = CB ? -b.y : b.y;
+    acc.x = fma(a.x, b.x, acc.x);
+    acc.x = fma(-aim, bim, acc.x);
+    acc.y = fma(a.x, bim, acc.y);
+    acc.y = fma(aim, b.x, acc.y);
+}
+template <bool CA, bool CB>
+__device__ inline void cmac(float& acc, float a, float b) { acc = fmaf(a, b, acc); }
+template <bool CA, bool CB>
+__device__ inline void cmac(double& acc, double a, double b) { acc = fma(a, b, acc); }
+
+template <typename Acc>
+__device__ inline Acc axpby(double alpha, Acc v, double beta, Acc c);
+template <> __device__ inline float axpby(double a, float v, double b, float c) {
+    return (float)(a * v + b * c);
+}
+template <> __device__ inline double axpby(double a, double v, double b, double c) {
+    return a * v + b * c;
+}
+template <> __device__ inline f2 axpby(double a, f2 v, double b, f2 c) {
+    return f2{(float)(a * v.x + b * c.x), (float)(a * v.y + b * c.y)};
+}
+template <> __device__ inline d2 axpby(double a, d2 v, double b, d2 c) {
+    return d2{a * v.x + b * c.x, a * v.y + b * c.y};
+}
+
+// Triangular block lift: t -> (bi, bj), bi >= bj, t = bi*(bi+1)/2 + bj.
+__device__ inline void lift_tri(long t, long& bi, long& bj) {
+    long i = (long)((sqrt(8.0 * (double)t + 1.0) - 1.0) * 0.5);
+    while ((i + 1) * (i + 2) / 2 <= t) ++i;
+    while (i * (i + 1) / 2 > t) --i;
+    bi = i;
+    bj = t - i * (i + 1) / 2;
+}
+
+/* --------------------- generic herk (fallback path) -------------------- */
+// C[b][i][j] (i>=j) = alpha * sum_k f(a(i,k), a(j,k)) + beta * C
+//   a(i,k) = a_base[b*a_bstride + i*a_n + k*a_k]   (element strides)
+//   f = conj(x)*y if CONJ_FIRST else x*conj(y); plain product for reals.
+template <typename Loader, typename Acc, bool CONJ_FIRST>
+__global__ void herk_generic_kernel(long n, long k, long nbatch,
+                                    double alpha, const void* a, long a_n,
+                                    long a_k, long a_b, double beta, void* c,
+                                    long c_row, long c_b, long ntiles) {
+    constexpr int TB = 16;
+    __shared__ Acc sa[TB][TB + 1];
+    __shared__ Acc sb[TB][TB + 1];
+    int tx = threadIdx.x;  // j within tile
+    int ty = threadIdx.y;  // i within tile
+    for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
+        const void* ab = a;
+        Acc* cb = (Acc*)c + batch * c_b;
+        long aoff = batch * a_b;
+        for (long t = blockIdx.x; t < ntiles; t += gridDim.x) {
+            long bi, bj;
+            lift_tri(t, bi, bj);
+            long i0 = bi * TB, j0 = bj * TB;
+            Acc acc{};
+            for (long k0 = 0; k0 < k; k0 += TB) {
+                // stage a(i0+ty, k0+tx) and a(j0+ty, k0+tx)
+                long ii = i0 + ty, jj = j0 + ty, kk = k0 + tx;
+                if (ii < n && kk < k)
+                    sa[ty][tx] = Loader::load(ab, aoff + ii * a_n + kk * a_k);
+                else
+                    sa[ty][tx] = Acc{};
+                if (jj < n && kk < k)
+                    sb[ty][tx] = Loader::load(ab, aoff + jj * a_n + kk * a_k);
+                else
+                    sb[ty][tx] = Acc{};
+                __syncthreads();
+                int klim = (int)min((long)TB, k - k0);
+                for (int q = 0; q < klim; ++q) {
+                    cmac<CONJ_FIRST, !CONJ_FIRST>(acc, sa[ty][q], sb[tx][q]);
+                }
+                __syncthreads();
+            }
+            long i = i0 + ty, j = j0 + tx;
+            if (i < n && j < n && i >= j) {
+                Acc prev = beta != 0.0 ? cb[i * c_row + j] : Acc{};
+                cb[i * c_row + j] = axpby<Acc>(alpha, acc, beta, prev);
+            }
+        }
+    }
+}
+
+/* ------------------ specialized correlator cherk (ci8) ------------------ */
+// Per-channel C = X^H.X: C[b][i][j] (i>=j) = alpha*sum_k conj(A[k,i])*A[k,j]
+// + beta*C, with A ci8 k-major: element (k,i) at a + b*a_b + k*lda + i.
+// 256 threads = 16x16 lanes of 2x2 outputs -> 32x32 tile; K staged in LDS
+// slabs of 32.  All loads dword (2 ci8 elements) — the dispatch guarantees
+// n, lda, a_b even (reference routing conditions, linalg.cu:210-226).
+template <int BK>
+__global__ void cherk_ci8_kernel(long n, long k, long nbatch, float alpha,
+                                 const signed char* __restrict__ a, long lda,
+                                 long a_b, float beta, f2* __restrict__ c,
+                                 long c_row, long c_b, long ntiles) {
+    // LDS: two slabs (i-strip, j-strip), [BK][32] ci8 elements as uint16
+    __shared__ short sa[BK][32 + 2];
+    __shared__ short sb[BK][32 + 2];
+    int lane = threadIdx.x;          // 0..255
+    int tj = lane & 15;              // j pair index
+    int ti = lane >> 4;              // i pair index
+    for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        for (long t = blockIdx.x; t < ntiles; t += gridDim.x) {
+            long bi, bj;
+            lift_tri(t, bi, bj);
+            long i0 = bi * 32, j0 = bj * 32;
+            bool diag = bi == bj;
+            f2 acc00{}, acc01{}, acc10{}, acc11{};
+            for (long k0 = 0; k0 < k; k0 += BK) {
+                // Stage: 256 threads load [BK][32] elements as dwords
+                // (2 elements each): 16 dwords per k-row, BK rows.
+                {
+                    int col2 = lane & 15;   // dword index within row (2 elems)
+                    int krow = lane >> 4;   // 16 rows per pass
+                    for (int kk = krow; kk < BK; kk += 16) {
+                        long kg = k0 + kk;
+                        int i = col2 * 2;
+                        unsigned v_i = 0, v_j = 0;
+                        if (kg < k) {
+                            if (i0 + i < n)
+                                v_i = *(const unsigned*)(ab + (kg * lda + i0 + i) * 2);
+                            if (j0 + i < n)
+                                v_j = *(const unsigned*)(ab + (kg * lda + j0 + i) * 2);
+                        }
+                        sa[kk][i] = (short)(v_i & 0xFFFF);
+                        sa[kk][i + 1] = (short)(v_i >> 16);
+                        sb[kk][i] = (short)(v_j & 0xFFFF);
+                        sb[kk][i + 1] = (short)(v_j >> 16);
+                    }
+                }
+                __syncthreads();
+                int klim = (int)min((long)BK, k - k0);
+                for (int q = 0; q < klim; ++q) {
+                    short ra0 = sa[q][2 * ti], ra1 = sa[q][2 * ti + 1];
+                    short rb0 = sb[q][2 * tj], rb1 = sb[q][2 * tj + 1];
+                    f2 a0{(float)(signed char)(ra0 & 0xFF), (float)(signed char)(ra0 >> 8)};
+                    f2 a1{(float)(signed char)(ra1 & 0xFF), (float)(signed char)(ra1 >> 8)};
+                    f2 b0{(float)(signed char)(rb0 & 0xFF), (float)(signed char)(rb0 >> 8)};
+                    f2 b1{(float)(signed char)(rb1 & 0xFF), (float)(signed char)(rb1 >> 8)};
+                    cmac<true, false>(acc00, a0, b0);
+                    cmac<true, false>(acc01, a0, b1);
+                    cmac<true, false>(acc10, a1, b0);
+                    cmac<true, false>(acc11, a1, b1);
+                }
+                __syncthreads();
+            }
+            // Write 2x2 outputs at (i0+2ti+{0,1}, j0+2tj+{0,1}), lower only.
+            long i = i0 + 2 * ti, j = j0 + 2 * tj;
+            f2 accs[2][2] = {{acc00, acc01}, {acc10, acc11}};
+            for (int di = 0; di < 2; ++di) {
+                for (int dj = 0; dj < 2; ++dj) {
+                    long ii = i + di, jj = j + dj;
+                    if (ii >= n || jj >= n) continue;
+                    if (diag && ii < jj) continue;
+                    f2 prev = beta != 0.f ? cb[ii * c_row + jj] : f2{};
+                    f2 v = accs[di][dj];
+                    cb[ii * c_row + jj] =
+                        f2{alpha * v.x + beta * prev.x,
+                           alpha * v.y + beta * prev.y};
+                }
+            }
+        }
+    }
+}
+
+/* ------------------------ generic gemm (fallback) ----------------------- */
+// C[b][i][j] = alpha * sum_k a^(i,k) * b^(k,j) + beta * C[b][i][j]
+template <typename LoadA, typename LoadB, typename Acc>
+__global__ void gemm_generic_kernel(long m, long nn, long k, long nbatch,
+                                    double alpha, const void* a, long a_i,
+                                    long a_k, long a_b, int conj_a,
+                                    const void* b, long b_k, long b_j,
+                                    long b_b, int conj_b, double beta,
+                                    void* c, long c_row, long c_b,
+                                    long ntiles, long tiles_j) {
+    constexpr int TB = 16;
+    __shared__ Acc sa[TB][TB + 1];
+    __shared__ Acc sb[TB][TB + 1];
+    int tx = threadIdx.x;
+    int ty = threadIdx.y;
+    for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
+        long aoff = batch * a_b;
+        long boff = batch * b_b;
+        Acc* cb = (Acc*)c + batch * c_b;
+        for (long t = blockIdx.x; t < ntiles; t += gridDim.x) {
+            long bi = t / tiles_j, bj = t % tiles_j;
+            long i0 = bi * TB, j0 = bj * TB;
+            Acc acc{};
+            for (long k0 = 0; k0 < k; k0 += TB) {
+                long ii = i0 + ty, kk = k0 + tx;
+                sa[ty][tx] = (ii < m && kk < k)
+                                 ? LoadA::load(a, aoff + ii * a_i + kk * a_k)
+                                 : Acc{};
+                long kk2 = k0 + ty, jj = j0 + tx;
+                sb[ty][tx] = (kk2 < k && jj < nn)
+                                 ? LoadB::load(b, boff + kk2 * b_k + jj * b_j)
+                                 : Acc{};
+                __syncthreads();
+                int klim = (int)min((long)TB, k - k0);
+                if (!conj_a && !conj_b) {
+                    for (int q = 0; q < klim; ++q)
+                        cmac<false, false>(acc, sa[ty][q], sb[q][tx]);
+                } else if (conj_a && !conj_b) {
+                    for (int q = 0; q < klim; ++q)
+                        cmac<true, false>(acc, sa[ty][q], sb[q][tx]);
+                } else if (!conj_a && conj_b) {
+                    for (int q = 0; q < klim; ++q)
+                        cmac<false, true>(acc, sa[ty][q], sb[q][tx]);
+                } else {
+                    for (int q = 0; q < klim; ++q)
+                        cmac<true, true>(acc, sa[ty][q], sb[q][tx]);
+                }
+                __syncthreads();
+            }
+            long i = i0 + ty, j = j0 + tx;
+            if (i < m && j < nn) {
+                Acc prev = beta != 0.0 ? cb[i * c_row + j] : Acc{};
+                cb[i * c_row + j] = axpby<Acc>(alpha, acc, beta, prev);
+            }
+        }
+    }
+}
+
+/* -------------------------- beamform kernel ----------------------------- */
+// C[b][i][j] = alpha*sum_k W(i,k)*X(j,k) + beta*C   (no conjugation)
+//   W: LoadW at w + b*w_b + i*ldw + k      (k-fastest)
+//   X: LoadX at x + b*x_b + j*ldx + k      (k-fastest)
+// MTILE beams per launch chunk kept in registers; W chunk cached in LDS.
+template <typename LoadW, typename LoadX, int MTILE>
+__global__ void beamform_kernel(long mm, long nn, long k, long nbatch,
+                                float alpha, const void* w, long ldw, long w_b,
+                                const void* x, long ldx, long x_b, float beta,
+                                f2* __restrict__ c, long c_row, long c_b,
+                                long i0) {
+    __shared__ f2 sw[MTILE][64 + 1];
+    int lane = threadIdx.x;  // 256 threads; each owns consecutive j
+    long mlim = min((long)MTILE, mm - i0);
+    for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
+        const void* wb = w;
+        const void* xb = x;
+        long woff = batch * w_b;
+        long xoff = batch * x_b;
+        f2* cb = c + batch * c_b;
+        for (long j0 = (long)blockIdx.x * blockDim.x; j0 < nn;
+             j0 += (long)gridDim.x * blockDim.x) {
+            long j = j0 + lane;
+            f2 acc[MTILE];
+            for (int m = 0; m < MTILE; ++m) acc[m] = f2{};
+            for (long k0 = 0; k0 < k; k0 += 64) {
+                int klim = (int)min((long)64, k - k0);
+                // stage W[i0..i0+mlim)[k0..k0+klim) into LDS
+                for (int idx = lane; idx < (int)mlim * klim; idx += 256) {
+                    int m = idx / klim, kk = idx % klim;
+                    sw[m][kk] = LoadW::load(wb, woff + (i0 + m) * ldw + k0 + kk);
+                }
+                __syncthreads();
+                if (j < nn) {
+                    for (int kk = 0; kk < klim; ++kk) {
+                        f2 xv = LoadX::load(xb, xoff + j * ldx + k0 + kk);
+                        for (int m = 0; m < MTILE; ++m) {
+                            if (m < mlim) cmac<false, false>(acc[m], sw[m][kk], xv);
+                        }
+                    }
+                }
+                __syncthreads();
+            }
+            if (j < nn) {
+                for (int m = 0; m < (int)mlim; ++m) {
+                    f2 prev = beta != 0.f ? cb[(i0 + m) * c_row + j] : f2{};
+                    cb[(i0 + m) * c_row + j] =
+                        f2{alpha * acc[m].x + beta * prev.x,
+                           alpha * acc[m].y + beta * prev.y};
+                }
+            }
+        }
+    }
+}
+
+/* ------------------------------ dispatch -------------------------------- */
+
+struct MatView {
+    const BFarray* arr;
+    long n_stride;   // element stride of the "row" index (i or n)
+    long k_stride;   // element stride of the contraction index
+    long batch_stride;
+    bool conj;
+};
+
+long elem_strides(const BFarray* a, long* es) {
+    int nb = bfamd::dtype_nbyte(a->dtype);
+    for (int d = 0; d < a->ndim; ++d) {
+        if (a->strides[d] % nb != 0) return -1;
+        es[d] = a->strides[d] / nb;
+    }
+    return nb;
+}
+
+unsigned cap_grid(long v, long cap) {
+    return (unsigned)std::min<long>(std::max<long>(v, 1), cap);
+}
+
+// herk dispatch: C = alpha * op(A) + beta*C over the lower triangle.
+// a_n/a_k/a_b element strides; conj_first selects conj placement.
+BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
+                     long nbatch, double alpha, const void* a, long a_n,
+                     long a_k, long a_b, double beta, void* c, long c_row,
+                     long c_b, bool conj_first, hipStream_t stream) {
+    // Specialized correlator kernel: ci8, k-major (a_n==1), conj-first,
+    // even n/lda/batch strides (dword loads).
+    if (a_type == BF_DTYPE_CI8 && c_type == BF_DTYPE_CF32 && conj_first &&
+        a_n == 1 && n % 2 == 0 && a_k % 2 == 0 && a_b % 2 == 0) {
+        long ntiles_dim = (n + 31) / 32;
+        long ntiles = ntiles_dim * (ntiles_dim + 1) / 2;
+        dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
+        hipLaunchKernelGGL((cherk_ci8_kernel<32>), grid, dim3(256), 0, stream,
+                           n, k, nbatch, (float)alpha,
+                           (const signed char*)a, a_k, a_b, (float)beta,
+                           (f2*)c, c_row, c_b, ntiles);
+        BF_CHECK_HIP(hipGetLastError());
+        return BF_STATUS_SUCCESS;
+    }
+    long ntiles_dim = (n + 15) / 16;
+    long ntiles = ntiles_dim * (ntiles_dim + 1) / 2;
+    dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
+    dim3 block(16, 16);
+#define HERK_CASE(LOADER, ACC)                                               \
+    do {                                                                     \
+        if (conj_first)                                                      \
+            hipLaunchKernelGGL((herk_generic_kernel<LOADER<ACC>, ACC, true>),\
+                               grid, block, 0, stream, n, k, nbatch, alpha,  \
+                               a, a_n, a_k, a_b, beta, c, c_row, c_b,        \
+                               ntiles);                                      \
+        else                                                                 \
+            hipLaunchKernelGGL((herk_generic_kernel<LOADER<ACC>, ACC, false>),\
+                               grid, block, 0, stream, n, k, nbatch, alpha,  \
+                               a, a_n, a_k, a_b, beta, c, c_row, c_b,        \
+                               ntiles);                                      \
+        BF_CHECK_HIP(hipGetLastError());                                     \
+        return BF_STATUS_SUCCESS;                                            \
+    } while (0)
+    switch (a_type) {
+        case BF_DTYPE_CI8:
+            BF_ASSERT(c_type == BF_DTYPE_CF32, BF_STATUS_UNSUPPORTED_DTYPE);
+            HERK_CASE(LoadCI8, f2);
+        case BF_DTYPE_CI16:
+            BF_ASSERT(c_type == BF_DTYPE_CF32, BF_STATUS_UNSUPPORTED_DTYPE);
+            HERK_CASE(LoadCI16, f2);
+        case BF_DTYPE_CF32:
+            BF_ASSERT(c_type == BF_DTYPE_CF32, BF_STATUS_UNSUPPORTED_DTYPE);
+            HERK_CASE(LoadCF32, f2);
+        case BF_DTYPE_CF64:
+            BF_ASSERT(c_type == BF_DTYPE_CF64, BF_STATUS_UNSUPPORTED_DTYPE);
+            HERK_CASE(LoadCF64, d2);
+        case BF_DTYPE_F32:
+            BF_ASSERT(c_type == BF_DTYPE_F32, BF_STATUS_UNSUPPORTED_DTYPE);
+            HERK_CASE(LoadF32, float);
+        case BF_DTYPE_F64:
+            BF_ASSERT(c_type == BF_DTYPE_F64, BF_STATUS_UNSUPPORTED_DTYPE);
+            HERK_CASE(LoadF64, double);
+        default:
+            return BF_STATUS_UNSUPPORTED_DTYPE;
+    }
+#undef HERK_CASE
+}
+
+BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
+                     long nn, long k, long nbatch, double alpha,
+                     const void* a, long a_i, long a_k, long a_b, bool conj_a,
+                     const void* b, long b_k, long b_j, long b_b, bool conj_b,
+                     double beta, void* c, long c_row, long c_b,
+                     hipStream_t stream) {
+    // Specialized beamformer: W (a) k-fast x X (b) k-fast, mixed int/float
+    // inputs, cf32 out, no conj (reference bf_cgemm_TN_smallM conditions,
+    // linalg.cu:665-678, generalized to any nbeam via 16-beam chunks).
+    if ((b_type == BF_DTYPE_CI8 || b_type == BF_DTYPE_CI4) &&
+        (a_type == BF_DTYPE_CI16 || a_type == BF_DTYPE_CF32) &&
+        c_type == BF_DTYPE_CF32 && a_k == 1 && b_k == 1 && !conj_a &&
+        !conj_b && m <= 1024) {
+        dim3 grid(cap_grid((nn + 255) / 256, 4096), cap_grid(nbatch, 65535));
+        for (long i0 = 0; i0 < m; i0 += 16) {
+#define BEAM_CASE(LW, LX)                                                     \
+    hipLaunchKernelGGL((beamform_kernel<LW<f2>, LX<f2>, 16>), grid,           \
+                       dim3(256), 0, stream, m, nn, k, nbatch, (float)alpha,  \
+                       a, a_i, a_b, b, b_j, b_b, (float)beta, (f2*)c, c_row,  \
+                       c_b, i0)
+            if (a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI8)
+                BEAM_CASE(LoadCF32, LoadCI8);
+            else if (a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI4)
+                BEAM_CASE(LoadCF32, LoadCI4);
+            else if (a_type == BF_DTYPE_CI16 && b_type == BF_DTYPE_CI8)
+                BEAM_CASE(LoadCI16, LoadCI8);
+            else
+                BEAM_CASE(LoadCI16, LoadCI4);
+#undef BEAM_CASE
+            BF_CHECK_HIP(hipGetLastError());
+        }
+        return BF_STATUS_SUCCESS;
+    }
+
+    long tiles_i = (m + 15) / 16, tiles_j = (nn + 15) / 16;
+    long ntiles = tiles_i * tiles_j;
+    dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
+    dim3 block(16, 16);
+#define GEMM_CASE(LA, LB, ACC)                                                \
+    do {                                                                      \
+        hipLaunchKernelGGL((gemm_generic_kernel<LA<ACC>, LB<ACC>, ACC>),      \
+                           grid, block, 0, stream, m, nn, k, nbatch, alpha,   \
+                           a, a_i, a_k, a_b, conj_a ? 1 : 0, b, b_k, b_j,     \
+                           b_b, conj_b ? 1 : 0, beta, c, c_row, c_b, ntiles,  \
+                           tiles_j);                                          \
+        BF_CHECK_HIP(hipGetLastError());                                      \
+        return BF_STATUS_SUCCESS;                                             \
+    } while (0)
+    if (a_type == b_type) {
+        switch (a_type) {
+            case BF_DTYPE_CI8:
+                BF_ASSERT(c_type == BF_DTYPE_CF32, BF_STATUS_UNSUPPORTED_DTYPE);
+                GEMM_CASE(LoadCI8, LoadCI8, f2);
+            case BF_DTYPE_CF32:
+                BF_ASSERT(c_type == BF_DTYPE_CF32, BF_STATUS_UNSUPPORTED_DTYPE);
+                GEMM_CASE(LoadCF32, LoadCF32, f2);
+            case BF_DTYPE_CF64:
+                BF_ASSERT(c_type == BF_DTYPE_CF64, BF_STATUS_UNSUPPORTED_DTYPE);
+                GEMM_CASE(LoadCF64, LoadCF64, d2);
+            case BF_DTYPE_F32:
+                BF_ASSERT(c_type == BF_DTYPE_F32, BF_STATUS_UNSUPPORTED_DTYPE);
+                GEMM_CASE(LoadF32, LoadF32, float);
+            case BF_DTYPE_F64:
+                BF_ASSERT(c_type == BF_DTYPE_F64, BF_STATUS_UNSUPPORTED_DTYPE);
+                GEMM_CASE(LoadF64, LoadF64, double);
+            default:
+                return BF_STATUS_UNSUPPORTED_DTYPE;
+        }
+    }
+    // Mixed dtypes outside the beamform fast-path conditions:
+    if ((a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI8) ||
+        (a_type == BF_DTYPE_CI8 && b_type == BF_DTYPE_CF32)) {
+        BF_ASSERT(c_type == BF_DTYPE_CF32, BF_STATUS_UNSUPPORTED_DTYPE);
+        if (a_type == BF_DTYPE_CF32) GEMM_CASE(LoadCF32, LoadCI8, f2);
+        GEMM_CASE(LoadCI8, LoadCF32, f2);
+    }
+    if ((a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI4))
+        GEMM_CASE(LoadCF32, LoadCI4, f2);
+    if ((a_type == BF_DTYPE_CI16 && b_type == BF_DTYPE_CI8))
+        GEMM_CASE(LoadCI16, LoadCI8, f2);
+    if ((a_type == BF_DTYPE_CI16 && b_type == BF_DTYPE_CI4))
+        GEMM_CASE(LoadCI16, LoadCI4, f2);
+    return BF_STATUS_UNSUPPORTED_DTYPE;
+#undef GEMM_CASE
+}
+
+/* -------------- batch-dim flattening + stride analysis ------------------ */
+
+struct BatchPlan {
+    long nbatch = 1;
+    int batch_dim = -1;
+    long batch_shape[BF_MAX_DIMS];
+    int ndim = 2;
+};
+
+// Shared batch analysis for 2 or 3 arrays (reference linalg.cu:272-316
+// semantics: flatten mergeable leading dims, keep last-2 (matrix) dims,
+// pick the largest remaining dim as the kernel batch dim, loop the rest).
+BFstatus plan_batches(const BFarray** arrs, BFarray* flat, int narr,
+                      BatchPlan* plan) {
+    int ndim = arrs[0]->ndim;
+    for (int i = 1; i < narr; ++i)
+        BF_ASSERT(arrs[i]->ndim == ndim, BF_STATUS_INVALID_SHAPE);
+    if (ndim <= 2) {
+        for (int i = 0; i < narr; ++i) flat[i] = *arrs[i];
+    } else {
+        unsigned long keep = 0x7ul << (ndim - 3);
+        for (int i = 0; i < narr; ++i) keep |= bfamd::padded_dims_mask(arrs[i]);
+        for (int i = 0; i < narr; ++i)
+            bfamd::flatten_dims(arrs[i], &flat[i], keep);
+    }
+    int fdim = flat[0].ndim;
+    for (int i = 1; i < narr; ++i)
+        BF_ASSERT(flat[i].ndim == fdim, BF_STATUS_INVALID_SHAPE);
+    plan->ndim = fdim;
+    for (int d = 0; d < fdim; ++d) plan->batch_shape[d] = 1;
+    plan->nbatch = 1;
+    plan->batch_dim = -1;
+    // c is the last array; batch dims follow c's shape
+    const BFarray& c = flat[narr - 1];
+    for (int d = 0; d < fdim - 2; ++d) {
+        for (int i = 0; i < narr - 1; ++i) {
+            BF_ASSERT(flat[i].shape[d] == c.shape[d] || flat[i].shape[d] == 1,
+                      BF_STATUS_INVALID_SHAPE);
+        }
+        plan->batch_shape[d] = c.shape[d];
+        if (c.shape[d] >= plan->nbatch) {
+            plan->nbatch = c.shape[d];
+            plan->batch_dim = d;
+        }
+    }
+    if (plan->batch_dim >= 0) plan->batch_shape[plan->batch_dim] = 1;
+    return BF_STATUS_SUCCESS;
+}
+
+// Iterate the residual batch dims (all but the kernel batch dim).
+template <typename F>
+BFstatus foreach_residual_batch(const BatchPlan& plan, F&& fn) {
+    long counters[BF_MAX_DIMS] = {0};
+    int nres = plan.ndim - 2;
+    for (;;) {
+        BF_CHECK(fn(counters));
+        int i = nres - 1;
+        for (; i >= 0; --i) {
+            if (++counters[i] < plan.batch_shape[i]) break;
+            counters[i] = 0;
+        }
+        if (i < 0) break;
+    }
+    return BF_STATUS_SUCCESS;
+}
+
+BFstatus matmul_aa(double alpha, const BFarray* a_in, bool adjoint,
+                   double beta, const BFarray* c_in) {
+    BF_ASSERT(c_in->ndim == a_in->ndim, BF_STATUS_INVALID_SHAPE);
+    BFarray a_adj = *a_in;
+    int nd = a_in->ndim;
+    if (adjoint) {
+        std::swap(a_adj.shape[nd - 1], a_adj.shape[nd - 2]);
+        std::swap(a_adj.strides[nd - 1], a_adj.strides[nd - 2]);
+        a_adj.conjugated = !a_adj.conjugated;
+    }
+    BF_ASSERT(c_in->shape[nd - 1] == a_adj.shape[nd - 2], BF_STATUS_INVALID_SHAPE);
+    BF_ASSERT(c_in->shape[nd - 2] == a_adj.shape[nd - 2], BF_STATUS_INVALID_SHAPE);
+
+    const BFarray* arrs[2] = {&a_adj, c_in};
+    BFarray flat[2];
+    BatchPlan plan;
+    BF_CHECK(plan_batches(arrs, flat, 2, &plan));
+    BFarray& a = flat[0];
+    BFarray& c = flat[1];
+    int ndim = plan.ndim;
+
+    long as[BF_MAX_DIMS], cs[BF_MAX_DIMS];
+    BF_ASSERT(elem_strides(&a, as) > 0, BF_STATUS_INVALID_STRIDE);
+    BF_ASSERT(elem_strides(&c, cs) > 0, BF_STATUS_INVALID_STRIDE);
+
+    // Stride analysis (reference linalg.cu:318-344 semantics):
+    //   n-index fastest (k-major layout) -> requires conjugated for complex
+    //     (the correlator layout); conj-first product.
+    //   k-index fastest (row-major [n,k]) -> requires NOT conjugated;
+    //     conj-second product.
+    long n = a.shape[ndim - 2], kk = a.shape[ndim - 1];
+    bool cplx = bfamd::dtype_is_complex(a.dtype);
+    long a_n, a_k;
+    bool conj_first;
+    if (as[ndim - 1] < as[ndim - 2]) {
+        // row-major [n,k]: fastest dim is k
+        BF_ASSERT(as[ndim - 1] == 1, BF_STATUS_UNSUPPORTED_STRIDE);
+        BF_ASSERT(!cplx || !a.conjugated, BF_STATUS_UNSUPPORTED);
+        a_n = as[ndim - 2];
+        a_k = 1;
+        conj_first = false;  // C[i][j] = sum a(i,k) conj(a(j,k))
+    } else if (as[ndim - 1] > as[ndim - 2]) {
+        // k-major [n,k] view: fastest dim is n
+        BF_ASSERT(as[ndim - 2] == 1, BF_STATUS_UNSUPPORTED_STRIDE);
+        BF_ASSERT(!cplx || a.conjugated, BF_STATUS_UNSUPPORTED);
+        a_n = 1;
+        a_k = as[ndim - 1];
+        conj_first = true;  // C[i][j] = sum conj(a(k,i)) a(k,j)
+    } else {
+        return BF_STATUS_INVALID_STRIDE;
+    }
+    BF_ASSERT(cs[ndim - 2] >= cs[ndim - 1], BF_STATUS_UNSUPPORTED_STRIDE);
+    long c_row = cs[ndim - 2];
+    long a_bs = plan.batch_dim >= 0
+                    ? (a.shape[plan.batch_dim] == 1 ? 0 : as[plan.batch_dim])
+                    : 0;
+    long c_bs = plan.batch_dim >= 0 ? cs[plan.batch_dim] : 0;
+
+    hipStream_t stream = bfamd::thread_stream();
+    int a_nbyte = bfamd::dtype_nbyte(a.dtype);
+    int c_nbyte = bfamd::dtype_nbyte(c.dtype);
+    return foreach_residual_batch(plan, [&](const long* counters) {
+        long aoff = 0, coff = 0;
+        for (int d = 0; d < ndim - 2; ++d) {
+            long ai = a.shape[d] == 1 ? 0 : counters[d];
+            aoff += ai * as[d];
+            coff += counters[d] * cs[d];
+        }
+        return launch_herk(a.dtype, c.dtype, n, kk, plan.nbatch, alpha,
+                           (const char*)a.data + aoff * a_nbyte, a_n, a_k,
+                           a_bs, beta, (char*)c.data + coff * c_nbyte, c_row,
+                           c_bs, conj_first, stream);
+    });
+}
+
+BFstatus matmul_ab(double alpha, const BFarray* a_in, const BFarray* b_in,
+                   double beta, const BFarray* c_in) {
+    int nd = a_in->ndim;
+    BF_ASSERT(c_in->ndim == nd && b_in->ndim == nd, BF_STATUS_INVALID_SHAPE);
+    BF_ASSERT(c_in->shape[nd - 2] == a_in->shape[nd - 2], BF_STATUS_INVALID_SHAPE);
+    BF_ASSERT(c_in->shape[nd - 1] == b_in->shape[nd - 1], BF_STATUS_INVALID_SHAPE);
+    BF_ASSERT(a_in->shape[nd - 1] == b_in->shape[nd - 2], BF_STATUS_INVALID_SHAPE);
+
+    const BFarray* arrs[3] = {a_in, b_in, c_in};
+    BFarray flat[3];
+    BatchPlan plan;
+    BF_CHECK(plan_batches(arrs, flat, 3, &plan));
+    BFarray &a = flat[0], &b = flat[1], &c = flat[2];
+    int ndim = plan.ndim;
+
+    long as[BF_MAX_DIMS], bs[BF_MAX_DIMS], cs[BF_MAX_DIMS];
+    BF_ASSERT(elem_strides(&a, as) > 0, BF_STATUS_INVALID_STRIDE);
+    BF_ASSERT(elem_strides(&b, bs) > 0, BF_STATUS_INVALID_STRIDE);
+    BF_ASSERT(elem_strides(&c, cs) > 0, BF_STATUS_INVALID_STRIDE);
+
+    long m = c.shape[ndim - 2], nn = c.shape[ndim - 1], kk = a.shape[ndim - 1];
+
+    // a: element (i, k); fastest dim determines layout.
+    long a_i, a_k;
+    bool conj_a = false;
+    bool a_cplx = bfamd::dtype_is_complex(a.dtype);
+    if (as[ndim - 1] < as[ndim - 2]) {
+        BF_ASSERT(as[ndim - 1] == 1, BF_STATUS_UNSUPPORTED_STRIDE);
+        BF_ASSERT(!a_cplx || !a.conjugated, BF_STATUS_UNSUPPORTED);
+        a_i = as[ndim - 2];
+        a_k = 1;
+    } else if (as[ndim - 1] > as[ndim - 2]) {
+        BF_ASSERT(as[ndim - 2] == 1, BF_STATUS_UNSUPPORTED_STRIDE);
+        conj_a = a_cplx && a.conjugated;
+        a_i = 1;
+        a_k = as[ndim - 1];
+    } else {
+        return BF_STATUS_INVALID_STRIDE;
+    }
+    // b: element (k, j)
+    long b_k, b_j;
+    bool conj_b = false;
+    bool b_cplx = bfamd::dtype_is_complex(b.dtype);
+    if (bs[ndim - 1] < bs[ndim - 2]) {
+        BF_ASSERT(bs[ndim - 1] == 1, BF_STATUS_UNSUPPORTED_STRIDE);
+        BF_ASSERT(!b_cplx || !b.conjugated, BF_STATUS_UNSUPPORTED);
+        b_k = bs[ndim - 2];
+        b_j = 1;
+    } else if (bs[ndim - 1] > bs[ndim - 2]) {
+        BF_ASSERT(bs[ndim - 2] == 1, BF_STATUS_UNSUPPORTED_STRIDE);
+        conj_b = b_cplx && b.conjugated;
+        b_k = 1;
+        b_j = bs[ndim - 1];
+    } else {
+        return BF_STATUS_INVALID_STRIDE;
+    }
+    BF_ASSERT(cs[ndim - 2] >= cs[ndim - 1], BF_STATUS_UNSUPPORTED_STRIDE);
+    long c_row = cs[ndim - 2];
+
+    long a_bs = plan.batch_dim >= 0
+                    ? (a.shape[plan.batch_dim] == 1 ? 0 : as[plan.batch_dim])
+                    : 0;
+    long b_bs = plan.batch_dim >= 0
+                    ? (b.shape[plan.batch_dim] == 1 ? 0 : bs[plan.batch_dim])
+                    : 0;
+    long c_bs = plan.batch_dim >= 0 ? cs[plan.batch_dim] : 0;
+
+    hipStream_t stream = bfamd::thread_stream();
+    int a_nbyte = bfamd::dtype_nbyte(a.dtype);
+    int b_nbyte = bfamd::dtype_nbyte(b.dtype);
+    int c_nbyte = bfamd::dtype_nbyte(c.dtype);
+    return foreach_residual_batch(plan, [&](const long* counters) {
+        long aoff = 0, boff = 0, coff = 0;
+        for (int d = 0; d < ndim - 2; ++d) {
+            aoff += (a.shape[d] == 1 ? 0 : counters[d]) * as[d];
+            boff += (b.shape[d] == 1 ? 0 : counters[d]) * bs[d];
+            coff += counters[d] * cs[d];
+        }
+        return launch_gemm(a.dtype, b.dtype, c.dtype, m, nn, kk, plan.nbatch,
+                           alpha, (const char*)a.data + aoff * a_nbyte, a_i,
+                           a_k, a_bs, conj_a,
+                           (const char*)b.data + boff * b_nbyte, b_k, b_j,
+                           b_bs, conj_b, beta,
+                           (char*)c.data + coff * c_nbyte, c_row, c_bs,
+                           stream);
+    });
+}
+
+}  // namespace
+
+struct BFlinalg_impl {
+    int dummy = 0;  // stateless on this backend (no BLAS handles)
+};
+
+extern "C" {
+
+BFstatus bfLinAlgCreate(BFlinalg* handle_ptr) {
+    BF_ASSERT(handle_ptr, BF_STATUS_INVALID_POINTER);
+    BF_TRY_RETURN({ *handle_ptr = new BFlinalg_impl(); });
+}
+
+BFstatus bfLinAlgDestroy(BFlinalg handle) {
+    BF_ASSERT(handle, BF_STATUS_INVALID_HANDLE);
+    delete handle;
+    return BF_STATUS_SUCCESS;
+}
+
+BFstatus bfLinAlgMatMul(BFlinalg handle, double alpha, BFarray const* a,
+                        BFarray const* b, double beta, BFarray const* c) {
+    using namespace bfamd;
+    BF_ASSERT(handle, BF_STATUS_INVALID_HANDLE);
+    BF_ASSERT(a || b, BF_STATUS_INVALID_ARGUMENT);
+    BF_ASSERT(c, BF_STATUS_INVALID_POINTER);
+    BF_ASSERT(space_device_accessible(c->space), BF_STATUS_UNSUPPORTED_SPACE);
+    if (a && b) {
+        BF_ASSERT(space_device_accessible(a->space), BF_STATUS_UNSUPPORTED_SPACE);
+        BF_ASSERT(space_device_accessible(b->space), BF_STATUS_UNSUPPORTED_SPACE);
+        return matmul_ab(alpha, a, b, beta, c);
+    }
+    const BFarray* input = a ? a : b;
+    BF_ASSERT(space_device_accessible(input->space), BF_STATUS_UNSUPPORTED_SPACE);
+    bool adjoint = (input == b);
+    return matmul_aa(alpha, input, adjoint, beta, c);
+}
+
+}  // extern "C"

@@ -1,0 +1,222 @@
+// bifrost_amd: bfTranspose — arbitrary-axes permutation copy on device
+// (feeder a6).  Behaviour contract: reference
+// src/transpose.cu:503-561 (device arrays only; element sizes 1..16 B;
+// exact byte permutation).  Implementation is our own, two kernels:
+//   1. rows kernel — fastest dim unmoved (the [t,c,sp]->[c,t,sp] feeder):
+//      each workgroup copies whole contiguous rows; coalesced both sides.
+//   2. tile kernel — fastest dim moves: LDS 32x32 tiles (+1 pad) over the
+//      (in-fastest, out-fastest) plane, batched over the other dims;
+//      coalesced reads AND writes.
+// Both are pure HBM streams (2x gulp bytes).
+
+#include <bifrost/transpose.h>
+
+#include <hip/hip_runtime.h>
+
+#include "dtype.hpp"
+#include "hipctx.hpp"
+#include "status.hpp"
+
+namespace {
+
+struct TransposeArgs {
+    int ndim;                    // output ndim
+    long oshape[BF_MAX_DIMS];    // output shape (elements)
+    long ostrides[BF_MAX_DIMS];  // output strides (elements of T)
+    long istrides[BF_MAX_DIMS];  // input stride for each OUTPUT dim (elements)
+};
+
+// Generic gather kernel: one thread per output element, coalesced writes.
+template <typename T>
+__global__ void transpose_gather_kernel(const T* __restrict__ in,
+                                        T* __restrict__ out,
+                                        TransposeArgs args, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        size_t rem = i;
+        long ioff = 0, ooff = 0;
+        for (int d = args.ndim - 1; d >= 0; --d) {
+            long idx = (long)(rem % (size_t)args.oshape[d]);
+            rem /= (size_t)args.oshape[d];
+            ioff += idx * args.istrides[d];
+            ooff += idx * args.ostrides[d];
+        }
+        out[ooff] = in[ioff];
+    }
+}
+
+// Rows kernel: output dims [0..ndim-2] permute, last dim is contiguous in
+// both input and output.  One workgroup per (row-batch), threads stream the
+// row.  rows = product(oshape[0..ndim-2]).
+template <typename T>
+__global__ void transpose_rows_kernel(const T* __restrict__ in,
+                                      T* __restrict__ out,
+                                      TransposeArgs args, long nrow,
+                                      long row_len) {
+    for (long row = blockIdx.x; row < nrow; row += gridDim.x) {
+        long rem = row;
+        long ioff = 0, ooff = 0;
+        for (int d = args.ndim - 2; d >= 0; --d) {
+            long idx = rem % args.oshape[d];
+            rem /= args.oshape[d];
+            ioff += idx * args.istrides[d];
+            ooff += idx * args.ostrides[d];
+        }
+        for (long j = threadIdx.x; j < row_len; j += blockDim.x) {
+            out[ooff + j] = in[ioff + j];
+        }
+    }
+}
+
+// 2D tile kernel: out-fastest dim (width W, input stride si_w) x the input-
+// fastest dim (height H, out stride so_h, input stride 1).  Batch over the
+// rest.  LDS 32x33 to dodge bank conflicts on the transposed read.
+template <typename T>
+__global__ void transpose_tile_kernel(const T* __restrict__ in,
+                                      T* __restrict__ out,
+                                      TransposeArgs args,
+                                      int wdim,  // output dim that is input-fastest
+                                      long W, long H,
+                                      long si_w, long so_h, long nbatch) {
+    __shared__ T tile[32][33];
+    // grid: (tilesW x tilesH, batch)
+    long tiles_w = (W + 31) / 32;
+    long tiles_h = (H + 31) / 32;
+    for (long tb = blockIdx.y; tb < nbatch; tb += gridDim.y) {
+        // batch offsets over dims other than ndim-1 and wdim
+        long rem = tb;
+        long ioff = 0, ooff = 0;
+        for (int d = args.ndim - 2; d >= 0; --d) {
+            if (d == wdim) continue;
+            long idx = rem % args.oshape[d];
+            rem /= args.oshape[d];
+            ioff += idx * args.istrides[d];
+            ooff += idx * args.ostrides[d];
+        }
+        for (long t = blockIdx.x; t < tiles_w * tiles_h; t += gridDim.x) {
+            long tw = t % tiles_w, th = t / tiles_w;
+            long w0 = tw * 32, h0 = th * 32;
+            // read: input row = h (contiguous in input), col = w
+            // in element (w, h) at ioff + w*si_w + h*1
+            {
+                int lw = threadIdx.x & 31;   // along input-fastest (h)
+                int lh = threadIdx.x >> 5;   // along w
+                for (int hh = lh; hh < 32; hh += (int)(blockDim.x >> 5)) {
+                    long w = w0 + hh, h = h0 + lw;
+                    if (w < W && h < H)
+                        tile[hh][lw] = in[ioff + w * si_w + h];
+                }
+            }
+            __syncthreads();
+            // write: output row = w (contiguous in output), col = h
+            // out element (w, h) at ooff + h*so_h + w*1
+            {
+                int lw = threadIdx.x & 31;   // along output-fastest (w)
+                int lh = threadIdx.x >> 5;
+                for (int hh = lh; hh < 32; hh += (int)(blockDim.x >> 5)) {
+                    long h = h0 + hh, w = w0 + lw;
+                    if (w < W && h < H)
+                        out[ooff + h * so_h + w] = tile[lw][hh];
+                }
+            }
+            __syncthreads();
+        }
+    }
+}
+
+template <typename T>
+BFstatus launch_transpose(const BFarray* in, const BFarray* out,
+                          const int* axes) {
+    using namespace bfamd;
+    int ndim = in->ndim;
+    int esize = (int)sizeof(T);
+    TransposeArgs args;
+    args.ndim = ndim;
+    size_t n = 1;
+    for (int d = 0; d < ndim; ++d) {
+        args.oshape[d] = out->shape[d];
+        args.ostrides[d] = out->strides[d] / esize;
+        args.istrides[d] = in->strides[axes[d]] / esize;
+        n *= (size_t)out->shape[d];
+    }
+    hipStream_t s = bfamd::thread_stream();
+    if (n == 0) return BF_STATUS_SUCCESS;
+
+    bool last_unmoved = axes[ndim - 1] == ndim - 1 &&
+                        args.istrides[ndim - 1] == 1 &&
+                        args.ostrides[ndim - 1] == 1;
+    if (ndim == 1 || last_unmoved) {
+        long row_len = args.oshape[ndim - 1];
+        long nrow = (long)(n / (size_t)row_len);
+        if (ndim == 1) { nrow = 1; }
+        unsigned blocks = (unsigned)std::min<long>(nrow > 0 ? nrow : 1, 65535);
+        // Wide rows want more threads; short rows want more blocks.
+        unsigned threads = row_len >= 256 ? 256 : 64;
+        hipLaunchKernelGGL(transpose_rows_kernel<T>, dim3(blocks),
+                           dim3(threads), 0, s, (const T*)in->data,
+                           (T*)out->data, args, nrow, row_len);
+        BF_CHECK_HIP(hipGetLastError());
+        return BF_STATUS_SUCCESS;
+    }
+
+    // Tiled case: output-fastest dim must be contiguous in output, and the
+    // input-fastest dim (axes^-1[ndim-1]) contiguous in input.
+    int wdim = -1;  // output dim whose input stride is 1
+    for (int d = 0; d < ndim; ++d)
+        if (args.istrides[d] == 1) wdim = d;
+    if (args.ostrides[ndim - 1] == 1 && wdim >= 0 && wdim != ndim - 1) {
+        long W = args.oshape[ndim - 1];      // output-fastest extent
+        long H = args.oshape[wdim];          // input-fastest extent
+        long si_w = args.istrides[ndim - 1];
+        long so_h = args.ostrides[wdim];
+        long nbatch = 1;
+        for (int d = 0; d < ndim - 1; ++d)
+            if (d != wdim) nbatch *= args.oshape[d];
+        long tiles = ((W + 31) / 32) * ((H + 31) / 32);
+        unsigned gx = (unsigned)std::min<long>(tiles, 8192);
+        unsigned gy = (unsigned)std::min<long>(nbatch, 65535);
+        hipLaunchKernelGGL(transpose_tile_kernel<T>, dim3(gx, gy), dim3(256),
+                           0, s, (const T*)in->data, (T*)out->data, args,
+                           wdim, W, H, si_w, so_h, nbatch);
+        BF_CHECK_HIP(hipGetLastError());
+        return BF_STATUS_SUCCESS;
+    }
+
+    // Fallback: generic gather (coalesced writes only).
+    unsigned blocks = (unsigned)std::min<size_t>((n + 255) / 256, 32768);
+    hipLaunchKernelGGL(transpose_gather_kernel<T>, dim3(blocks), dim3(256), 0,
+                       s, (const T*)in->data, (T*)out->data, args, n);
+    BF_CHECK_HIP(hipGetLastError());
+    return BF_STATUS_SUCCESS;
+}
+
+struct alignas(16) byte16 { unsigned long long a, b; };
+
+}  // namespace
+
+extern "C" BFstatus bfTranspose(BFarray const* in, BFarray const* out,
+                                int const* axes) {
+    using namespace bfamd;
+    BF_ASSERT(in && out && axes, BF_STATUS_INVALID_POINTER);
+    BF_ASSERT(!out->immutable, BF_STATUS_INVALID_POINTER);
+    BF_ASSERT(in->ndim == out->ndim, BF_STATUS_INVALID_SHAPE);
+    BF_ASSERT(in->ndim >= 1 && in->ndim <= BF_MAX_DIMS, BF_STATUS_INVALID_SHAPE);
+    for (int d = 0; d < in->ndim; ++d) {
+        BF_ASSERT(axes[d] >= 0 && axes[d] < in->ndim, BF_STATUS_INVALID_ARGUMENT);
+        BF_ASSERT(out->shape[d] == in->shape[axes[d]], BF_STATUS_INVALID_SHAPE);
+    }
+    // Device arrays only (matches reference transpose.cu:512).
+    BF_ASSERT(space_device_accessible(in->space) &&
+              space_device_accessible(out->space), BF_STATUS_UNSUPPORTED_SPACE);
+    int esize = dtype_nbyte(in->dtype);
+    BF_ASSERT(esize == dtype_nbyte(out->dtype), BF_STATUS_UNSUPPORTED_DTYPE);
+    switch (esize) {
+        case 1:  return launch_transpose<unsigned char>(in, out, axes);
+        case 2:  return launch_transpose<unsigned short>(in, out, axes);
+        case 4:  return launch_transpose<unsigned>(in, out, axes);
+        case 8:  return launch_transpose<unsigned long long>(in, out, axes);
+        case 16: return launch_transpose<byte16>(in, out, axes);
+        default: return BF_STATUS_UNSUPPORTED_DTYPE;
+    }
+}
