@@ -1,0 +1,201 @@
+"""GPU parity tests for bfLinAlgMatMul — the reference's own test matrix
+(test/test_linalg.py) re-run against the numpy oracle, on the HIP path.
+
+Shape sweeps pin the edge-case branches (odd ntime = K tail, nstand up to
+65 = partial tiles + diagonal handling, misalign = unaligned base), per
+SURVEY.md §4.  Tolerances are the reference's (RTOL=1e-4, ATOL=1e-5;
+correlator kernel 10x RTOL, test_linalg.py:41-42,185).
+"""
+
+import numpy as np
+import pytest
+
+import bifrost_amd as bf
+from bifrost_amd.linalg import LinAlg
+from oracle.linalg import H
+
+pytestmark = pytest.mark.gpu
+
+RTOL = 1e-4
+ATOL = 1e-5
+
+
+@pytest.fixture(scope="module")
+def linalg():
+    return LinAlg()
+
+
+def run_corr(linalg, ntime, nstand, nchan, misalign=0, beta=0.0):
+    np.random.seed(1234)
+    x_shape = (ntime, nchan, nstand * 2)
+    x8 = ((np.random.random(size=x_shape + (2,)) * 2 - 1) * 127).astype(np.int8)
+    x = x8.astype(np.float32).view(np.complex64).reshape(x_shape)
+    x = x.transpose(1, 0, 2)[..., misalign:]
+    b_gold = np.matmul(H(x), x)
+    triu = np.triu_indices(x.shape[-1], 1)
+    b_gold[..., triu[0], triu[1]] = 0
+    xb = bf.ndarray(x8.view(bf.DataType.ci8).reshape(x_shape))
+    xb = bf.asarray(xb, space="cuda")
+    xb = xb.transpose(1, 0, 2)[..., misalign:]
+    b = bf.zeros_like(b_gold, space="cuda")
+    linalg.matmul(1, None, xb, beta, b)
+    if beta:
+        b_gold = b_gold * (1 + beta)  # second accumulation identical
+        linalg.matmul(1, None, xb, beta, b)
+    b = b.copy("system")
+    np.testing.assert_allclose(np.asarray(b), b_gold, RTOL * 10, ATOL)
+
+
+def run_beam(linalg, ntime, nbeam, nstand, nchan):
+    np.random.seed(1234)
+    x_shape = (ntime, nchan, nstand * 2)
+    w_shape = (nbeam, nchan, nstand * 2)
+    x8 = ((np.random.random(size=x_shape + (2,)) * 2 - 1) * 127).astype(np.int8)
+    x = x8.astype(np.float32).view(np.complex64).reshape(x_shape)
+    w = ((np.random.random(size=w_shape + (2,)) * 2 - 1) * 127).astype(np.int8) \
+        .astype(np.float32).view(np.complex64).reshape(w_shape)
+    b_gold = np.matmul(w.transpose(1, 0, 2), x.transpose(1, 2, 0))
+    xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8).reshape(x_shape)),
+                    space="cuda")
+    wb = bf.asarray(w, space="cuda")
+    b = bf.zeros_like(b_gold, space="cuda")
+    linalg.matmul(1, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0), 0, b)
+    b = b.copy("system")
+    np.testing.assert_allclose(np.asarray(b), b_gold, RTOL, ATOL)
+
+
+class TestCorrelatorKernel:
+    def test_small_sweep(self, linalg):
+        for nchan in (1, 3, 5):
+            for ntime in (1, 2, 3, 4, 8, 12):
+                for nstand in (1, 2, 3, 4, 5, 8, 16, 17, 32, 33, 64, 65):
+                    for misalign in range(0, min(2 * (nstand - 1), 3), 2):
+                        run_corr(linalg, ntime, nstand, nchan, misalign)
+
+    def test_large(self, linalg):
+        run_corr(linalg, 100, 200, 1)
+        run_corr(linalg, 99, 200, 3)
+        run_corr(linalg, 400, 100, 7)
+        run_corr(linalg, 36, 97, 31)
+        run_corr(linalg, 4, 512, 1)
+        run_corr(linalg, 512, 256, 3)
+        run_corr(linalg, 1000, 256, 1)
+
+    def test_beta_accumulate(self, linalg):
+        # beta=1 accumulation across gulps (reference TODO; correlate.py:85
+        # depends on it)
+        run_corr(linalg, 64, 16, 3, beta=1.0)
+
+
+class TestBeamformerKernel:
+    def test_small_sweep(self, linalg):
+        for nchan in (1, 3):
+            for ntime in (1, 2, 3, 8):
+                for nstand in (16, 64, 256):
+                    for nbeam in (1, 2, 3, 7, 12):
+                        run_beam(linalg, ntime, nbeam, nstand, nchan)
+
+    def test_large(self, linalg):
+        for nbeam in (1, 5, 12):
+            run_beam(linalg, 512, nbeam, 256, 10)
+
+    def test_many_beams(self, linalg):
+        # beyond the reference's 16-beam cap (config 5 needs 64)
+        for nbeam in (16, 17, 33, 64):
+            run_beam(linalg, 64, nbeam, 64, 3)
+
+
+class TestMatMulAA:
+    def _run_shape(self, linalg, shape, dtype, axes=None, conj=False):
+        np.random.seed(1234)
+        a = ((np.random.random(size=shape)) * 127).astype(dtype)
+        if axes is None:
+            axes = list(range(len(shape)))
+        aa = a.transpose(axes)
+        if conj:
+            aa = aa.conj()
+        c_gold = np.matmul(aa, H(aa))
+        triu = np.triu_indices(shape[axes[-2]], 1)
+        c_gold[..., triu[0], triu[1]] = 0
+        ab = bf.asarray(a, space="cuda")
+        aab = ab.transpose(axes)
+        if conj:
+            aab = aab.conj()
+        c = bf.zeros_like(c_gold, space="cuda")
+        linalg.matmul(1, aab, None, 0, c)
+        c = c.copy("system")
+        np.testing.assert_allclose(np.asarray(c), c_gold, RTOL, ATOL)
+
+    @pytest.mark.parametrize("dtype", [np.float32, np.float64, np.complex64,
+                                       np.complex128])
+    def test_dtypes(self, linalg, dtype):
+        self._run_shape(linalg, (11, 23), dtype)
+        self._run_shape(linalg, (111, 223), dtype)
+        self._run_shape(linalg, (111, 223), dtype, [1, 0], conj=True)
+        self._run_shape(linalg, (3, 111, 223), dtype)
+        self._run_shape(linalg, (3, 111, 223), dtype, [0, 2, 1], conj=True)
+        self._run_shape(linalg, (3, 111, 223), dtype, [1, 0, 2])
+        self._run_shape(linalg, (5, 3, 111, 57), dtype)
+        self._run_shape(linalg, (5, 3, 111, 57), dtype, [1, 0, 2, 3])
+        self._run_shape(linalg, (5, 3, 111, 57), dtype, [0, 1, 3, 2], conj=True)
+
+    def test_ci8(self, linalg):
+        for transpose in (False, True):
+            for shape in [(11, 4), (12, 4), (11, 23), (111, 223), (3, 111, 222),
+                          (5, 3, 112, 224)]:
+                self._run_ci8(linalg, shape, transpose)
+
+    def _run_ci8(self, linalg, shape, transpose):
+        np.random.seed(1234)
+        shape_complex = shape[:-1] + (shape[-1] * 2,)
+        a8 = ((np.random.random(size=shape_complex) * 2 - 1) * 127).astype(np.int8)
+        a_gold = a8.astype(np.float32).view(np.complex64)
+        if transpose:
+            a_gold = H(a_gold)
+        c_gold = np.matmul(a_gold, H(a_gold))
+        triu = np.triu_indices(shape[-2] if not transpose else shape[-1], 1)
+        c_gold[..., triu[0], triu[1]] = 0
+        a = bf.asarray(bf.ndarray(a8.view(bf.DataType.ci8)), space="cuda")
+        if transpose:
+            a = a.transpose(
+                list(range(len(shape) - 2)) + [len(shape) - 1, len(shape) - 2]
+            ).conj()
+        c = bf.zeros_like(c_gold, space="cuda")
+        linalg.matmul(1, a, None, 0, c)
+        c = c.copy("system")
+        np.testing.assert_allclose(np.asarray(c), c_gold, RTOL, ATOL)
+
+
+class TestMatMulAB:
+    def _run(self, linalg, shape, k, dtype):
+        np.random.seed(1234)
+        ashape = shape[:-2] + (shape[-2], k)
+        bshape = shape[:-2] + (k, shape[-1])
+        a = ((np.random.random(size=ashape)) * 127).astype(dtype)
+        b = ((np.random.random(size=bshape)) * 127).astype(dtype)
+        c_gold = np.matmul(a, b)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.asarray(b, space="cuda")
+        c = bf.zeros_like(c_gold, space="cuda")
+        linalg.matmul(1, ag, bg, 0, c)
+        c = c.copy("system")
+        np.testing.assert_allclose(np.asarray(c), c_gold, RTOL, ATOL)
+
+    @pytest.mark.parametrize("dtype", [np.float32, np.float64, np.complex64,
+                                       np.complex128])
+    def test_dtypes(self, linalg, dtype):
+        self._run(linalg, (11, 23), 7, dtype)
+        self._run(linalg, (111, 223), 77, dtype)
+        self._run(linalg, (3, 111, 223), 77, dtype)
+
+    def test_transposed(self, linalg):
+        np.random.seed(1234)
+        a = ((np.random.random(size=(64, 32))) * 2 - 1).astype(np.complex64)
+        b = ((np.random.random(size=(32, 48))) * 2 - 1).astype(np.complex64)
+        c_gold = np.matmul(H(b), H(a))
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.asarray(b, space="cuda")
+        c = bf.zeros_like(c_gold, space="cuda")
+        linalg.matmul(1, H(bg), H(ag), 0, c)
+        np.testing.assert_allclose(np.asarray(c.copy("system")), c_gold,
+                                   RTOL, ATOL)
