@@ -197,6 +197,174 @@ __global__ void herk_generic_kernel(long n, long k, long nbatch,
     }
 }
 
+/* -------------- i8-MFMA correlator cherk (the v2 hot kernel) ------------ */
+// Per-channel C = X^H.X on mfma_i32_16x16x64_i8 tiles.  Complex arithmetic
+// decomposes into 4 planar int8 GEMMs with exact i32 accumulation:
+//   Re C[i][j] = Rr + Ri = sum ar_i ar_j + sum ai_i ai_j
+//   Im C[i][j] = X1 - X2 = sum ar_i ai_j - sum ai_i ar_j
+// Layout pipeline (probed on hardware, csrc/probe_mfma.hip):
+//   - global ci8 k-major rows -> deinterleave into planar re/im LDS tiles
+//     [colblock][k][16] (16-byte k-rows);
+//   - ds_read_tr8_b64 with per-lane addr = base + lane*8 delivers each lane
+//     the 16-lane-group's column transpose = exactly the A and B fragment
+//     layouts (lane l&15 = row/col, 16 k-bytes per lane over 2 reads);
+//   - the A/B k-permutation cancels as long as both operands use the same
+//     image (verified: any consistent partition passes).
+// Workgroup: 256 threads / 4 waves; 64x64 complex output tile (each wave a
+// 32x32 quadrant = 2x2 MFMA tiles x 4 planar accumulators = 64 AGPRs);
+// K slabs of 64 staged double-buffer-free v2 (optimization headroom noted
+// in DESIGN.md).  Requires n%2==0, 8-byte aligned rows (dispatch falls back
+// to the VALU kernel otherwise).
+
+typedef int v4i __attribute__((ext_vector_type(4)));
+typedef int v2i __attribute__((ext_vector_type(2)));
+typedef __attribute__((address_space(3))) v2i* lds_v2i;
+
+#define CHERK_BK 64
+
+__global__ __launch_bounds__(256)
+void cherk_ci8_mfma_kernel(long n, long k, long nbatch, float alpha,
+                           const signed char* __restrict__ a, long lda,
+                           long a_b, float beta, f2* __restrict__ c,
+                           long c_row, long c_b, long ntiles) {
+    // LDS planar images: [strip(i=0,j=1)][plane(re=0,im=1)][colblock 4][BK][16]
+    __shared__ signed char lds[2][2][4][CHERK_BK][16];
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;  // wave quadrant in the 64x64 tile
+
+    for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        for (long t = blockIdx.x; t < ntiles; t += gridDim.x) {
+            long bi, bj;
+            lift_tri(t, bi, bj);
+            long i0 = bi * 64, j0 = bj * 64;
+            bool diag = bi == bj;
+            v4i acc[2][2][4];  // [ti][tj][plane: rr, ii, ri, ir]
+            for (int x = 0; x < 2; ++x)
+                for (int y = 0; y < 2; ++y)
+                    for (int p = 0; p < 4; ++p) acc[x][y][p] = v4i{};
+
+            for (long k0 = 0; k0 < k; k0 += CHERK_BK) {
+                // ---- stage both strips into planar LDS tiles ----
+                // thread t: k-row kk = t>>2, col quarter q = t&3 (16 cols).
+                {
+                    int kk = tid >> 2, q = tid & 3;
+                    long kg = k0 + kk;
+                    for (int strip = 0; strip < 2; ++strip) {
+                        long base_col = strip ? j0 : i0;
+                        unsigned* ldre = (unsigned*)&lds[strip][0][q][kk][0];
+                        unsigned* ldim = (unsigned*)&lds[strip][1][q][kk][0];
+                        if (kg >= k) {
+                            for (int d = 0; d < 4; ++d) { ldre[d] = 0; ldim[d] = 0; }
+                            continue;
+                        }
+                        const signed char* row =
+                            ab + (kg * lda + base_col + 16 * q) * 2;
+                        for (int h = 0; h < 2; ++h) {
+                            long cbase = base_col + 16 * q + 8 * h;
+                            if (cbase + 8 <= n) {
+                                // 8 cols = 16 interleaved bytes = 4 dwords;
+                                // deinterleave to 2 re + 2 im dwords.
+                                const unsigned* rw = (const unsigned*)(row + 16 * h);
+                                unsigned d0 = rw[0], d1 = rw[1];
+                                unsigned d2 = rw[2], d3 = rw[3];
+                                unsigned re01 = (d0 & 0xFF) | ((d0 >> 8) & 0xFF00);
+                                unsigned im01 = ((d0 >> 8) & 0xFF) | ((d0 >> 16) & 0xFF00);
+                                unsigned re23 = (d1 & 0xFF) | ((d1 >> 8) & 0xFF00);
+                                unsigned im23 = ((d1 >> 8) & 0xFF) | ((d1 >> 16) & 0xFF00);
+                                unsigned re45 = (d2 & 0xFF) | ((d2 >> 8) & 0xFF00);
+                                unsigned im45 = ((d2 >> 8) & 0xFF) | ((d2 >> 16) & 0xFF00);
+                                unsigned re67 = (d3 & 0xFF) | ((d3 >> 8) & 0xFF00);
+                                unsigned im67 = ((d3 >> 8) & 0xFF) | ((d3 >> 16) & 0xFF00);
+                                ldre[2 * h] = re01 | (re23 << 16);
+                                ldim[2 * h] = im01 | (im23 << 16);
+                                ldre[2 * h + 1] = re45 | (re67 << 16);
+                                ldim[2 * h + 1] = im45 | (im67 << 16);
+                            } else {
+                                // edge tile: per-element with zero padding
+                                unsigned rev = 0, imv = 0;
+                                for (int e = 0; e < 8; ++e) {
+                                    signed char rr = 0, ii = 0;
+                                    if (cbase + e < n) {
+                                        rr = row[(8 * h + e) * 2];
+                                        ii = row[(8 * h + e) * 2 + 1];
+                                    }
+                                    rev |= (unsigned)(unsigned char)rr << (8 * (e & 3));
+                                    imv |= (unsigned)(unsigned char)ii << (8 * (e & 3));
+                                    if ((e & 3) == 3) {
+                                        ldre[2 * h + (e >> 2)] = rev;
+                                        ldim[2 * h + (e >> 2)] = imv;
+                                        rev = imv = 0;
+                                    }
+                                }
+                            }
+                        }
+                    }
+                }
+                __syncthreads();
+                // ---- MFMA over the slab ----
+                // Fragments: per (strip, plane, colblock): 16 B/lane via two
+                // tr8 reads at base + lane*8 and base + 512 + lane*8.
+                v4i fragI[2][2], fragJ[2][2];  // [plane][tile]
+                for (int p = 0; p < 2; ++p) {
+                    for (int ti = 0; ti < 2; ++ti) {
+                        int cb_i = 2 * wr + ti;
+                        const signed char* baseI = &lds[0][p][cb_i][0][0];
+                        v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                            (lds_v2i)(baseI + lane * 8));
+                        v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                            (lds_v2i)(baseI + 512 + lane * 8));
+                        fragI[p][ti] = v4i{lo[0], lo[1], hi[0], hi[1]};
+                        int cb_j = 2 * wc + ti;
+                        const signed char* baseJ = &lds[1][p][cb_j][0][0];
+                        lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                            (lds_v2i)(baseJ + lane * 8));
+                        hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                            (lds_v2i)(baseJ + 512 + lane * 8));
+                        fragJ[p][ti] = v4i{lo[0], lo[1], hi[0], hi[1]};
+                    }
+                }
+                for (int ti = 0; ti < 2; ++ti) {
+                    for (int tj = 0; tj < 2; ++tj) {
+                        acc[ti][tj][0] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                            fragI[0][ti], fragJ[0][tj], acc[ti][tj][0], 0, 0, 0);
+                        acc[ti][tj][1] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                            fragI[1][ti], fragJ[1][tj], acc[ti][tj][1], 0, 0, 0);
+                        acc[ti][tj][2] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                            fragI[0][ti], fragJ[1][tj], acc[ti][tj][2], 0, 0, 0);
+                        acc[ti][tj][3] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                            fragI[1][ti], fragJ[0][tj], acc[ti][tj][3], 0, 0, 0);
+                    }
+                }
+                __syncthreads();
+            }
+            // ---- epilogue: combine planes, scale, write lower triangle ----
+            // C/D layout: col = lane&15, row = (lane>>4)*4 + r.
+            for (int ti = 0; ti < 2; ++ti) {
+                for (int tj = 0; tj < 2; ++tj) {
+                    long ib = i0 + 32 * wr + 16 * ti;
+                    long jb = j0 + 32 * wc + 16 * tj;
+                    for (int r = 0; r < 4; ++r) {
+                        long i = ib + (lane >> 4) * 4 + r;
+                        long j = jb + (lane & 15);
+                        if (i >= n || j >= n) continue;
+                        if (diag && i < j) continue;
+                        if (i < j) continue;  // only lower triangle stored
+                        float re = (float)(acc[ti][tj][0][r] + acc[ti][tj][1][r]);
+                        float im = (float)(acc[ti][tj][2][r] - acc[ti][tj][3][r]);
+                        f2 prev = beta != 0.f ? cb[i * c_row + j] : f2{};
+                        cb[i * c_row + j] = f2{alpha * re + beta * prev.x,
+                                               alpha * im + beta * prev.y};
+                    }
+                }
+            }
+        }
+    }
+}
+
 /* ------------------ specialized correlator cherk (ci8) ------------------ */
 // Per-channel C = X^H.X: C[b][i][j] (i>=j) = alpha*sum_k conj(A[k,i])*A[k,j]
 // + beta*C, with A ci8 k-major: element (k,i) at a + b*a_b + k*lda + i.
@@ -422,10 +590,23 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                      long nbatch, double alpha, const void* a, long a_n,
                      long a_k, long a_b, double beta, void* c, long c_row,
                      long c_b, bool conj_first, hipStream_t stream) {
-    // Specialized correlator kernel: ci8, k-major (a_n==1), conj-first,
-    // even n/lda/batch strides (dword loads).
+    // Specialized correlator kernels: ci8, k-major (a_n==1), conj-first,
+    // even n/lda/batch strides (dword loads).  The i8-MFMA kernel is the
+    // product path; the VALU kernel remains as the odd-alignment fallback.
     if (a_type == BF_DTYPE_CI8 && c_type == BF_DTYPE_CF32 && conj_first &&
         a_n == 1 && n % 2 == 0 && a_k % 2 == 0 && a_b % 2 == 0) {
+        const char* disable = getenv("BIFROST_NO_MFMA");
+        if (!(disable && atoi(disable))) {
+            long ntiles_dim = (n + 63) / 64;
+            long ntiles = ntiles_dim * (ntiles_dim + 1) / 2;
+            dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
+            hipLaunchKernelGGL(cherk_ci8_mfma_kernel, grid, dim3(256), 0,
+                               stream, n, k, nbatch, (float)alpha,
+                               (const signed char*)a, a_k, a_b, (float)beta,
+                               (f2*)c, c_row, c_b, ntiles);
+            BF_CHECK_HIP(hipGetLastError());
+            return BF_STATUS_SUCCESS;
+        }
         long ntiles_dim = (n + 31) / 32;
         long ntiles = ntiles_dim * (ntiles_dim + 1) / 2;
         dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
