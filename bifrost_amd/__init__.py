@@ -8,7 +8,12 @@ see INTEGRATION.md).
 
 __version__ = "0.1.0"
 
-from bifrost_amd import device  # noqa: F401
+from bifrost_amd import affinity, device, memory  # noqa: F401
+from bifrost_amd import pipeline  # noqa: F401
+from bifrost_amd.pipeline import Pipeline, block_scope, get_default_pipeline  # noqa: F401
+from bifrost_amd.ring2 import Ring  # noqa: F401
+from bifrost_amd import blocks  # noqa: F401
+from bifrost_amd.block_chainer import BlockChainer  # noqa: F401
 from bifrost_amd.DataType import DataType  # noqa: F401
 from bifrost_amd.ndarray import (asarray, copy_array, empty, empty_like,  # noqa: F401
                                  memset_array, ndarray, zeros, zeros_like)

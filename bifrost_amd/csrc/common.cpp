@@ -25,6 +25,20 @@ hipStream_t& thread_stream() {
     return s;
 }
 
+// CPU-only hosts (no visible GPU) still run the system-space paths; stream
+// sync becomes a no-op there instead of a device error.
+bool hip_available() {
+    static int count = [] {
+        int c = 0;
+        if (hipGetDeviceCount(&c) != hipSuccess) {
+            (void)hipGetLastError();
+            return 0;
+        }
+        return c;
+    }();
+    return count > 0;
+}
+
 }  // namespace bfamd
 
 extern "C" {
@@ -80,6 +94,7 @@ BFstatus bfStreamSet(void const* stream) {
 }
 
 BFstatus bfStreamSynchronize(void) {
+    if (!bfamd::hip_available()) return BF_STATUS_SUCCESS;
     BF_CHECK_HIP(hipStreamSynchronize(bfamd::thread_stream()));
     return BF_STATUS_SUCCESS;
 }
