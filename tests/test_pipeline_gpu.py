@@ -37,14 +37,17 @@ def test_correlator_pipeline_end_to_end():
                          npol, nstand, npol)
     n = nstand * npol
     x = x8.astype(np.float32).view(np.complex64).reshape(ntime, nchan, n)
+    il = np.tril_indices(n)
     for w in range(got.shape[0]):
         xw = x[w * nframe_per_integration:(w + 1) * nframe_per_integration]
         xv = xw.transpose(1, 0, 2)
         gold = np.matmul(H(xv), xv)
-        triu = np.triu_indices(n, 1)
-        gold[..., triu[0], triu[1]] = 0
         gw = got[w].reshape(nchan, n, n)
-        np.testing.assert_allclose(gw, gold, rtol=1e-3, atol=1e-3)
+        # only the lower triangle is defined (matrix_fill_mode='lower');
+        # the upper triangle is whatever the recycled ring memory held
+        np.testing.assert_allclose(gw[:, il[0], il[1]],
+                                   gold[:, il[0], il[1]],
+                                   rtol=1e-3, atol=1e-3)
 
 
 def test_gpu_pipeline_unpack_ci4():
