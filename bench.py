@@ -161,7 +161,8 @@ def main():
     def step(i):
         beta = 0.0 if (i % GULPS_PER_INTEGRATION == 0) else 1.0
         linalg.matmul(1, None, x_view, beta, vis)
-        if args.time_split and (i + 1) % GULPS_PER_INTEGRATION == 0:
+        if (args.time_split and distributed and
+                (i + 1) % GULPS_PER_INTEGRATION == 0):
             import torch.distributed as dist
             dist.all_reduce(vis_t)
 

@@ -31,6 +31,7 @@
 #include <algorithm>
 #include <cmath>
 #include <cstring>
+#include <cstdint>
 
 #include "dtype.hpp"
 #include "hipctx.hpp"
@@ -430,6 +431,285 @@ void cherk_ci8_mfma_kernel(long n, long k, long nbatch, float alpha,
     }
 }
 
+/* ------- register-staged cooperative i8-MFMA cherk (T14/G15 form) ------- */
+// The binding resource at this tile shape is operand INGEST, not MFMA: the
+// LDS-DMA (global_load_lds) path caps at ~12 B/cyc/CU while the tile needs
+// ~50 B/cyc/CU at MFMA-bound pace.  This variant stages through registers
+// instead: plain dwordx4 loads (L2-resident re-reads stream at ~56
+// B/cyc/CU) into 4 VGPR quads per lane, then ds_write_b128 into the
+// chunk-swizzled [k][128] image.  Double-buffered, ONE bare s_barrier per
+// K-slab (no glds in flight, so __syncthreads carries no vmcnt drain);
+// loads for slab s+2 issue before computing slab s and fly across the
+// barrier (T14 write-after-barrier schedule).  Cooperative staging halves
+// the bytes vs the wave-private kernel (each strip staged once per
+// workgroup).  Full 16-B-aligned tiles only; edge shapes use the
+// cooperative glds kernel.
+template <int NHALF>  // K-slab = 64*NHALF (barriers amortize with NHALF)
+__global__ __launch_bounds__(256)
+void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
+                              const signed char* __restrict__ a, long lda,
+                              long a_b, float beta, f2* __restrict__ c,
+                              long c_row, long c_b, long ntiles) {
+    __shared__ signed char lds[2][2][64 * NHALF][128];  // [buf][strip]
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;
+
+    int tr_row = 8 * (lane >> 4) + ((lane & 15) >> 1);
+    int tr_half = lane & 1;
+
+    // staging: 128 threads per strip; thread covers 64 contiguous global
+    // bytes of one k-row (row tt>>1, byte half 64*(tt&1)).
+    int st_strip = tid >> 7;
+    int tt = tid & 127;
+    int st_row = tt >> 1;
+    int st_cq = (tt & 1) * 4;  // first of 4 consecutive 16-B chunks
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        long bi, bj;
+        lift_tri(t, bi, bj);
+        long i0 = bi * 64, j0 = bj * 64;
+        v4i acc[4][4];
+        for (int x = 0; x < 4; ++x)
+            for (int y = 0; y < 4; ++y) acc[x][y] = v4i{};
+
+        long base_col = st_strip ? j0 : i0;
+        const long slab_step = (64 * NHALF) * lda * 2;
+        const long half_step = 64 * lda * 2;
+        const signed char* src0 = ab + (long)st_row * lda * 2 +
+                                  base_col * 2 + 64 * (tt & 1);
+        v4i stg[4 * NHALF];
+        auto load_slab = [&](long s) {
+            const signed char* p = src0 + s * slab_step;
+            for (int h = 0; h < NHALF; ++h) {
+                const v4i* pv = (const v4i*)(p + h * half_step);
+                stg[4 * h + 0] = pv[0];
+                stg[4 * h + 1] = pv[1];
+                stg[4 * h + 2] = pv[2];
+                stg[4 * h + 3] = pv[3];
+            }
+        };
+        auto write_slab = [&](int buf) {
+            int swz = st_row & 7;
+            for (int h = 0; h < NHALF; ++h) {
+                signed char* base = &lds[buf][st_strip][64 * h + st_row][0];
+                *(v4i*)(base + 16 * ((st_cq + 0) ^ swz)) = stg[4 * h + 0];
+                *(v4i*)(base + 16 * ((st_cq + 1) ^ swz)) = stg[4 * h + 1];
+                *(v4i*)(base + 16 * ((st_cq + 2) ^ swz)) = stg[4 * h + 2];
+                *(v4i*)(base + 16 * ((st_cq + 3) ^ swz)) = stg[4 * h + 3];
+            }
+        };
+        auto frag = [&](const signed char* base, int cc) {
+            const signed char* p = base + tr_row * 128 +
+                16 * (cc ^ (tr_row & 7)) + 8 * tr_half;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 32 * 128));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto compute = [&](int buf, int h) {
+            const signed char* bI = &lds[buf][0][64 * h][0];
+            const signed char* bJ = &lds[buf][1][64 * h][0];
+            v4i fa[4], fb[4];
+            for (int ta = 0; ta < 4; ++ta) {
+                fa[ta] = frag(bI, 4 * wr + ta);
+                fb[ta] = frag(bJ, 4 * wc + ta);
+            }
+            for (int ta = 0; ta < 4; ++ta)
+                for (int tb = 0; tb < 4; ++tb)
+                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                        fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+        };
+
+        int nslab = (int)(k / (64 * NHALF));
+        load_slab(0);
+        write_slab(0);
+        if (nslab > 1) load_slab(1);
+        __syncthreads();
+        int buf = 0;
+        for (int s = 0; s < nslab; ++s) {
+            // write slab s+1 into the other buffer (its loads were issued
+            // one compute-phase ago), then issue loads for slab s+2 so
+            // they fly across the compute and the bare barrier.
+            if (s + 1 < nslab) {
+                write_slab(buf ^ 1);
+                if (s + 2 < nslab) load_slab(s + 2);
+            }
+            for (int h = 0; h < NHALF; ++h) compute(buf, h);
+            __syncthreads();
+            buf ^= 1;
+        }
+        for (int ta = 0; ta < 4; ++ta) {
+            for (int tb = 0; tb < 4; ++tb) {
+                long arow0 = i0 + 32 * wr + 8 * ta;
+                long acol = j0 + 32 * wc + 8 * tb + ((lane & 15) >> 1);
+                for (int p = 0; p < 2; ++p) {
+                    int v0 = acc[ta][tb][2 * p];
+                    int v1 = acc[ta][tb][2 * p + 1];
+                    int sv0 = __shfl_xor(v0, 1);
+                    int sv1 = __shfl_xor(v1, 1);
+                    long i = arow0 + 2 * (lane >> 4) + p;
+                    long j = acol;
+                    bool write = (lane & 1) == 0 && i < n && j < n && i >= j;
+                    if (write) {
+                        float re = (float)(v0 + sv1);
+                        float im = (float)(sv0 - v1);
+                        f2 prev = beta != 0.f ? cb[i * c_row + j] : f2{};
+                        cb[i * c_row + j] = f2{alpha * re + beta * prev.x,
+                                               alpha * im + beta * prev.y};
+                    }
+                }
+            }
+        }
+    }
+}
+
+/* ---------- wave-autonomous i8-MFMA cherk (no workgroup barriers) -------- */
+// Each wave stages ITS OWN operand halves (the 64-byte column half of the
+// i-strip and of the j-strip its quadrant consumes) into wave-private LDS,
+// double-buffered, ordered purely by per-wave counted s_waitcnt vmcnt —
+// there is no __syncthreads in the K loop at all, so waves on one SIMD
+// never stall each other at barriers.  The duplicated staging (each strip
+// half is loaded by two waves) is served by the XCD-local L2 (see the
+// channel mapping note below).  Full tiles only: n%64==0, k%64==0, k>=128;
+// edge shapes use the cooperative kernel.
+// Chunk swizzle: 16-B chunk c of row r lives at position c ^ swz(r),
+// swz(r) = (r ^ (r>>2)) & 3 — spreads the tr8 gather (row stride 64 B)
+// over distinct banks for all 16 source addresses of a lane group.
+__global__ __launch_bounds__(256)
+void cherk_ci8_mfma_wave_kernel(long n, long k, long nbatch, float alpha,
+                                const signed char* __restrict__ a, long lda,
+                                long a_b, float beta, f2* __restrict__ c,
+                                long c_row, long c_b, long ntiles) {
+    // [wave][buf][strip(i,j)][64 rows][64 B]  = 64 KB total
+    __shared__ signed char lds[4][2][2][64][64];
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;
+
+    int tr_row = 8 * (lane >> 4) + ((lane & 15) >> 1);
+    int tr_half = lane & 1;
+    int tr_swz_base = (tr_row ^ (tr_row >> 2)) & 3;
+
+    // staging lane map: row = lane>>2 (16 rows/glds), chunk = lane&2bits
+    int st_row = lane >> 2;
+    int st_chunk = (lane & 3) ^ ((st_row ^ (st_row >> 2)) & 3);
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        long bi, bj;
+        lift_tri(t, bi, bj);
+        long i0 = bi * 64, j0 = bj * 64;
+        bool diag = bi == bj;
+        v4i acc[4][4];
+        for (int x = 0; x < 4; ++x)
+            for (int y = 0; y < 4; ++y) acc[x][y] = v4i{};
+
+        // per-lane source pointers for the two halves this wave stages
+        // half I: complex cols [i0 + 32*wr, +32); half J: [j0 + 32*wc, +32)
+        const long row_step = 16 * lda * 2;        // 16 k-rows per glds pass
+        const signed char* srcI = ab + (long)st_row * lda * 2 +
+            (i0 + 32 * wr) * 2 + 16 * st_chunk;
+        const signed char* srcJ = ab + (long)st_row * lda * 2 +
+            (j0 + 32 * wc) * 2 + 16 * st_chunk;
+        signed char* dstI0 = &lds[wave][0][0][0][0];
+        signed char* dstJ0 = &lds[wave][0][1][0][0];
+        const long slab_step = CHERK_BK * lda * 2;
+
+        auto stage = [&](int buf, long slab) {
+            const signed char* sI = srcI + slab * slab_step;
+            const signed char* sJ = srcJ + slab * slab_step;
+            signed char* dI = dstI0 + buf * (64 * 64 * 2);
+            signed char* dJ = dstJ0 + buf * (64 * 64 * 2);
+            for (int p = 0; p < 4; ++p) {
+                __builtin_amdgcn_global_load_lds(
+                    (glob_u32)(sI + p * row_step),
+                    (lds_u32)(dI + p * 1024), 16, 0, 0);
+                __builtin_amdgcn_global_load_lds(
+                    (glob_u32)(sJ + p * row_step),
+                    (lds_u32)(dJ + p * 1024), 16, 0, 0);
+            }
+        };
+        // fragment gather: tile ta (16-byte col block of this wave's half)
+        auto frag = [&](const signed char* base, int ta) {
+            const signed char* p = base + tr_row * 64 +
+                16 * (ta ^ tr_swz_base) + 8 * tr_half;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 32 * 64));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto compute = [&](int buf) {
+            const signed char* bI = dstI0 + buf * (64 * 64 * 2);
+            const signed char* bJ = dstJ0 + buf * (64 * 64 * 2);
+            v4i fa[4], fb[4];
+            for (int ta = 0; ta < 4; ++ta) {
+                fa[ta] = frag(bI, ta);
+                fb[ta] = frag(bJ, ta);
+            }
+            for (int ta = 0; ta < 4; ++ta)
+                for (int tb = 0; tb < 4; ++tb)
+                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                        fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+        };
+
+        int nslab = (int)(k / CHERK_BK);
+        stage(0, 0);
+        stage(1, 1);
+        for (int s = 0; s < nslab; ++s) {
+            if (s + 1 < nslab)
+                asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            // fragment reads for slab s (lgkm waits inserted by compiler
+            // before the MFMAs); stage s+2 may overwrite buf s%2 only
+            // after these LDS reads issue+complete -- the MFMA operand
+            // waits order them, and the overwrite lands later (vm queue).
+            compute(s % 2);
+            if (s + 2 < nslab) stage((s + 2) % 2, s + 2);
+        }
+        // epilogue identical to the cooperative kernel
+        for (int ta = 0; ta < 4; ++ta) {
+            for (int tb = 0; tb < 4; ++tb) {
+                long arow0 = i0 + 32 * wr + 8 * ta;
+                long acol = j0 + 32 * wc + 8 * tb + ((lane & 15) >> 1);
+                for (int p = 0; p < 2; ++p) {
+                    int v0 = acc[ta][tb][2 * p];
+                    int v1 = acc[ta][tb][2 * p + 1];
+                    int sv0 = __shfl_xor(v0, 1);
+                    int sv1 = __shfl_xor(v1, 1);
+                    long i = arow0 + 2 * (lane >> 4) + p;
+                    long j = acol;
+                    bool write = (lane & 1) == 0 && i < n && j < n && i >= j;
+                    (void)diag;
+                    if (write) {
+                        float re = (float)(v0 + sv1);
+                        float im = (float)(sv0 - v1);
+                        f2 prev = beta != 0.f ? cb[i * c_row + j] : f2{};
+                        cb[i * c_row + j] = f2{alpha * re + beta * prev.x,
+                                               alpha * im + beta * prev.y};
+                    }
+                }
+            }
+        }
+    }
+}
+
 /* ------------------ specialized correlator cherk (ci8) ------------------ */
 // Per-channel C = X^H.X: C[b][i][j] (i>=j) = alpha*sum_k conj(A[k,i])*A[k,j]
 // + beta*C, with A ci8 k-major: element (k,i) at a + b*a_b + k*lda + i.
@@ -669,12 +949,46 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
             // participates in the remap.
             long nflat = ((ntiles * nbatch + 7) / 8) * 8;
             dim3 grid(cap_grid(nflat, 65535), 1);
-            // A 3-buffer counted-vmcnt pipeline exists (PIPE=true) but
-            // measures ~12% slower than the plain double buffer at equal
-            // occupancy on this kernel; keep it available for later
-            // schedule work via BIFROST_CHERK_PIPE=1.
-            const char* pipe_env = getenv("BIFROST_CHERK_PIPE");
-            bool pipe = pipe_env && atoi(pipe_env) &&
+            // Kernel selection: the wave-autonomous kernel (no workgroup
+            // barriers, per-wave counted vmcnt) for aligned full-tile
+            // shapes; the cooperative kernel for edges.  BIFROST_CHERK=coop
+            // forces the cooperative kernel; =pipe adds the 3-buffer
+            // workgroup pipeline (kept for schedule experiments).
+            const char* sel = getenv("BIFROST_CHERK");
+            bool aligned = (n % 64 == 0) && (k % 64 == 0) && k >= 128;
+            // register-staged kernel needs 16-B-aligned dwordx4 loads
+            bool al16 = aligned && ((uintptr_t)a % 16 == 0) &&
+                        ((a_k * 2) % 16 == 0) && ((a_b * 2) % 16 == 0);
+            bool want_rs = !sel || strcmp(sel, "rs") == 0;
+            bool want_wave = sel && strcmp(sel, "wave") == 0;
+            if (al16 && want_rs) {
+                const char* bkenv = getenv("BIFROST_CHERK_BK");
+                int nhalf = bkenv ? atoi(bkenv) / 64 : 2;
+                if (nhalf >= 2 && k % 128 == 0)
+                    hipLaunchKernelGGL(cherk_ci8_mfma_rs_kernel<2>, grid,
+                                       dim3(256), 0, stream, n, k, nbatch,
+                                       (float)alpha, (const signed char*)a,
+                                       a_k, a_b, (float)beta, (f2*)c, c_row,
+                                       c_b, ntiles);
+                else
+                    hipLaunchKernelGGL(cherk_ci8_mfma_rs_kernel<1>, grid,
+                                       dim3(256), 0, stream, n, k, nbatch,
+                                       (float)alpha, (const signed char*)a,
+                                       a_k, a_b, (float)beta, (f2*)c, c_row,
+                                       c_b, ntiles);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
+            if (aligned && (want_wave || (!sel || strcmp(sel, "coop") != 0))) {
+                hipLaunchKernelGGL(cherk_ci8_mfma_wave_kernel, grid,
+                                   dim3(256), 0, stream, n, k, nbatch,
+                                   (float)alpha, (const signed char*)a, a_k,
+                                   a_b, (float)beta, (f2*)c, c_row, c_b,
+                                   ntiles);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
+            bool pipe = sel && strcmp(sel, "pipe") == 0 &&
                         (n % 64 == 0) && (k % 128 == 0) && k >= 128;
             if (pipe)
                 hipLaunchKernelGGL(cherk_ci8_mfma_kernel<true>, grid,
