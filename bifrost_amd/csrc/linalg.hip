@@ -487,10 +487,13 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
         const signed char* src0 = ab + (long)st_row * lda * 2 +
                                   base_col * 2 + 64 * (tt & 1);
         v4i stg[4 * NHALF];
-        auto load_slab = [&](long s) {
-            const signed char* p = src0 + s * slab_step;
+        const signed char* load_next = src0;  // strength-reduced cursor
+        auto load_slab = [&]() {
+            const signed char* p = load_next;
+            load_next += slab_step;
             for (int h = 0; h < NHALF; ++h) {
-                const v4i* pv = (const v4i*)(p + h * half_step);
+                const v4i* pv = (const v4i*)__builtin_assume_aligned(
+                    p + h * half_step, 16);
                 stg[4 * h + 0] = pv[0];
                 stg[4 * h + 1] = pv[1];
                 stg[4 * h + 2] = pv[2];
@@ -530,9 +533,9 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
         };
 
         int nslab = (int)(k / (64 * NHALF));
-        load_slab(0);
+        load_slab();
         write_slab(0);
-        if (nslab > 1) load_slab(1);
+        if (nslab > 1) load_slab();
         __syncthreads();
         int buf = 0;
         for (int s = 0; s < nslab; ++s) {
@@ -541,7 +544,7 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
             // they fly across the compute and the bare barrier.
             if (s + 1 < nslab) {
                 write_slab(buf ^ 1);
-                if (s + 2 < nslab) load_slab(s + 2);
+                if (s + 2 < nslab) load_slab();
             }
             for (int h = 0; h < NHALF; ++h) compute(buf, h);
             __syncthreads();
