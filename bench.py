@@ -202,13 +202,18 @@ def main():
     per_launch_s = (kernel_ms / 1e3) / args.steps
     alg_bytes = algorithmic_bytes_per_step(ntime, nchan, N)
     achieved_gbs = alg_bytes / per_launch_s / 1e9
+    # Measured HBM/fabric traffic per launch comes from separate rocprofv3
+    # --pmc passes (TCC_EA0_RDREQ_sum x 64 x 2 per the gfx950 FETCH_SIZE
+    # calibration; see profiles/).  A wrapper that ran the PMC pass exports
+    # it here; the plain run reports null.
+    traffic_env = os.environ.get("BIFROST_TRAFFIC_BYTES_PER_LAUNCH")
     roofline = {
         "bound": "hbm",
         "achieved": round(achieved_gbs, 1),
         "peak": HBM_PEAK_GBS,
         "unit": "GB/s",
         "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
-        "traffic": None,
+        "traffic": float(traffic_env) if traffic_env else None,
     }
 
     if rank == 0:
