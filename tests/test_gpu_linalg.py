@@ -104,6 +104,32 @@ class TestBeamformerKernel:
         for nbeam in (16, 17, 33, 64):
             run_beam(linalg, 64, nbeam, 64, 3)
 
+    def _run_ci4(self, linalg, ntime, nbeam, nstand, nchan):
+        # 4-bit voltages (config 5): packed per Complex<FourBit> — re in
+        # the HIGH nibble (src/Complex.hpp:149-168)
+        np.random.seed(1234)
+        ks = nstand * 2
+        xi = np.random.randint(-7, 8, size=(ntime, nchan, ks, 2))
+        packed = (((xi[..., 0] & 0xF) << 4) | (xi[..., 1] & 0xF)) \
+            .astype(np.uint8)
+        x = (xi[..., 0] + 1j * xi[..., 1]).astype(np.complex64)
+        w = ((np.random.random((nbeam, nchan, ks, 2)) * 2 - 1) * 127) \
+            .astype(np.int8).astype(np.float32).view(np.complex64) \
+            .reshape(nbeam, nchan, ks)
+        b_gold = np.matmul(w.transpose(1, 0, 2), x.transpose(1, 2, 0))
+        xb = bf.asarray(bf.ndarray(packed.view(bf.DataType.ci4)),
+                        space="cuda")
+        wb = bf.asarray(w, space="cuda")
+        b = bf.zeros_like(b_gold, space="cuda")
+        linalg.matmul(1, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0), 0, b)
+        np.testing.assert_allclose(np.asarray(b.copy("system")), b_gold,
+                                   RTOL, ATOL)
+
+    def test_ci4_input(self, linalg):
+        for (t, b, s, c) in [(16, 3, 16, 2), (64, 12, 32, 3),
+                             (64, 64, 128, 2), (33, 7, 16, 1)]:
+            self._run_ci4(linalg, t, b, s, c)
+
 
 class TestMatMulAA:
     def _run_shape(self, linalg, shape, dtype, axes=None, conj=False):
