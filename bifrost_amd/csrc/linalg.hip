@@ -444,8 +444,12 @@ void cherk_ci8_mfma_kernel(long n, long k, long nbatch, float alpha,
 // the bytes vs the wave-private kernel (each strip staged once per
 // workgroup).  Full 16-B-aligned tiles only; edge shapes use the
 // cooperative glds kernel.
+#ifndef CHERK_RS_W1
+#define CHERK_RS_W1 4
+#endif
 template <int NHALF>  // K-slab = 64*NHALF (barriers amortize with NHALF)
 __global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(NHALF == 1 ? CHERK_RS_W1 : 2)))
 void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
                               const signed char* __restrict__ a, long lda,
                               long a_b, float beta, f2* __restrict__ c,
@@ -966,7 +970,9 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
             bool want_wave = sel && strcmp(sel, "wave") == 0;
             if (al16 && want_rs) {
                 const char* bkenv = getenv("BIFROST_CHERK_BK");
-                int nhalf = bkenv ? atoi(bkenv) / 64 : 2;
+                // BK=64 at 3 waves/SIMD beats BK=128 at 2 (occupancy wins
+                // over barrier amortization; measured 1.19 vs 1.11 Gsamp/s)
+                int nhalf = bkenv ? atoi(bkenv) / 64 : 1;
                 if (nhalf >= 2 && k % 128 == 0)
                     hipLaunchKernelGGL(cherk_ci8_mfma_rs_kernel<2>, grid,
                                        dim3(256), 0, stream, n, k, nbatch,
