@@ -481,6 +481,11 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
         long bi, bj;
         lift_tri(t, bi, bj);
         long i0 = bi * 64, j0 = bj * 64;
+        bool diag = bi == bj;
+        // On diagonal tiles the (wr<wc) wave quadrant lies entirely above
+        // the diagonal and skips its MFMA work (it still stages/syncs);
+        // in the wr==wc quadrant, tiles ta<tb are skipped too.
+        bool skip_all = diag && wr < wc;
         v4i acc[4][4];
         for (int x = 0; x < 4; ++x)
             for (int y = 0; y < 4; ++y) acc[x][y] = v4i{};
@@ -523,6 +528,7 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
             return v4i{lo[0], lo[1], hi[0], hi[1]};
         };
         auto compute = [&](int buf, int h) {
+            if (skip_all) return;
             const signed char* bI = &lds[buf][0][64 * h][0];
             const signed char* bJ = &lds[buf][1][64 * h][0];
             v4i fa[4], fb[4];
