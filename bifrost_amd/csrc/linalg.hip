@@ -723,6 +723,166 @@ void cherk_ci8_mfma_wave_kernel(long n, long k, long nbatch, float alpha,
     }
 }
 
+/* -------- 8-wave rectangular-tile register-staged cherk (rs8) ----------- */
+// 512 threads / 8 waves per 128x64-complex output tile (wave grid 4x2,
+// each wave the same 32x32-complex quadrant as the rs kernel).  Staged
+// bytes per MFMA drop 25% vs the square 64x64 tile (the i-strip is shared
+// by twice as many waves), attacking the measured ~39% staging share.
+// Same register profile as rs (acc 4x4 v4i folded into the unified file,
+// amdgpu_waves_per_eu(4)); LDS 48 KB -> 2 workgroups of 8 waves per CU at
+// 4 waves/SIMD.  I-strip [64][256] uses a 4-bit chunk swizzle (row stride
+// 256 B = a full bank row, so cc ^= r&15); J-strip [64][128] keeps the
+// 3-bit swizzle.  Requires n%128==0, k%64==0, k>=128, 16-B alignment.
+__global__ __launch_bounds__(512)
+__attribute__((amdgpu_waves_per_eu(4)))
+void cherk_ci8_mfma_rs8_kernel(long n, long k, long nbatch, float alpha,
+                               const signed char* __restrict__ a, long lda,
+                               long a_b, float beta, f2* __restrict__ c,
+                               long c_row, long c_b, long nti, long ntiles) {
+    __shared__ signed char ldsI[2][CHERK_BK][256];  // [buf][row][256 B]
+    __shared__ signed char ldsJ[2][CHERK_BK][128];
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;  // 4x2 wave grid
+
+    int tr_row = 8 * (lane >> 4) + ((lane & 15) >> 1);
+    int tr_half = lane & 1;
+
+    // staging: threads 0-255 stage the I strip (64 B per thread), threads
+    // 256-511 the J strip (32 B per thread); row = (tid&255)>>2.
+    int st_isJ = tid >> 8;
+    int st_row = (tid & 255) >> 2;
+    int st_q = tid & 3;
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        // rectangular lower-triangle tile map: row block I (128 complex),
+        // col block J (64 complex), J <= 2I+1.
+        long I = 0, rem = t;
+        while (rem >= (2 * I + 2 < nti ? 2 * I + 2 : nti)) {
+            rem -= (2 * I + 2 < nti ? 2 * I + 2 : nti);
+            ++I;
+        }
+        long J = rem;
+        long i0 = I * 128, j0 = J * 64;
+        // wave quadrant entirely above the diagonal?
+        bool skip_all = (i0 + 32 * wr + 31) < (j0 + 32 * wc);
+        v4i acc[4][4];
+        for (int x = 0; x < 4; ++x)
+            for (int y = 0; y < 4; ++y) acc[x][y] = v4i{};
+
+        const long slab_step = CHERK_BK * lda * 2;
+        const signed char* src0 = st_isJ
+            ? ab + (long)st_row * lda * 2 + j0 * 2 + 32 * st_q
+            : ab + (long)st_row * lda * 2 + i0 * 2 + 64 * st_q;
+        const signed char* load_next = src0;
+        v4i stg[4];
+        auto load_slab = [&]() {
+            const signed char* p = load_next;
+            load_next += slab_step;
+            if (st_isJ) {
+                const v4i* pv = (const v4i*)__builtin_assume_aligned(p, 16);
+                stg[0] = pv[0];
+                stg[1] = pv[1];
+            } else {
+                const v4i* pv = (const v4i*)__builtin_assume_aligned(p, 16);
+                stg[0] = pv[0]; stg[1] = pv[1];
+                stg[2] = pv[2]; stg[3] = pv[3];
+            }
+        };
+        auto write_slab = [&](int buf) {
+            if (st_isJ) {
+                signed char* base = &ldsJ[buf][st_row][0];
+                int swz = st_row & 7;
+                int c0 = 2 * st_q;
+                *(v4i*)(base + 16 * ((c0 + 0) ^ swz)) = stg[0];
+                *(v4i*)(base + 16 * ((c0 + 1) ^ swz)) = stg[1];
+            } else {
+                signed char* base = &ldsI[buf][st_row][0];
+                int swz = st_row & 15;
+                int c0 = 4 * st_q;
+                *(v4i*)(base + 16 * ((c0 + 0) ^ swz)) = stg[0];
+                *(v4i*)(base + 16 * ((c0 + 1) ^ swz)) = stg[1];
+                *(v4i*)(base + 16 * ((c0 + 2) ^ swz)) = stg[2];
+                *(v4i*)(base + 16 * ((c0 + 3) ^ swz)) = stg[3];
+            }
+        };
+        auto fragI = [&](int buf, int cc) {
+            const signed char* p = &ldsI[buf][0][0] + tr_row * 256 +
+                16 * (cc ^ (tr_row & 15)) + 8 * tr_half;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 32 * 256));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto fragJ = [&](int buf, int cc) {
+            const signed char* p = &ldsJ[buf][0][0] + tr_row * 128 +
+                16 * (cc ^ (tr_row & 7)) + 8 * tr_half;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 32 * 128));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto compute = [&](int buf) {
+            if (skip_all) return;
+            v4i fa[4], fb[4];
+            for (int ta = 0; ta < 4; ++ta) {
+                fa[ta] = fragI(buf, 4 * wr + ta);
+                fb[ta] = fragJ(buf, 4 * wc + ta);
+            }
+            for (int ta = 0; ta < 4; ++ta)
+                for (int tb = 0; tb < 4; ++tb)
+                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                        fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+        };
+
+        int nslab = (int)(k / CHERK_BK);
+        load_slab();
+        write_slab(0);
+        if (nslab > 1) load_slab();
+        __syncthreads();
+        int buf = 0;
+        for (int s = 0; s < nslab; ++s) {
+            if (s + 1 < nslab) {
+                write_slab(buf ^ 1);
+                if (s + 2 < nslab) load_slab();
+            }
+            compute(buf);
+            __syncthreads();
+            buf ^= 1;
+        }
+        for (int ta = 0; ta < 4; ++ta) {
+            for (int tb = 0; tb < 4; ++tb) {
+                long arow0 = i0 + 32 * wr + 8 * ta;
+                long acol = j0 + 32 * wc + 8 * tb + ((lane & 15) >> 1);
+                for (int p = 0; p < 2; ++p) {
+                    int v0 = acc[ta][tb][2 * p];
+                    int v1 = acc[ta][tb][2 * p + 1];
+                    int sv0 = __shfl_xor(v0, 1);
+                    int sv1 = __shfl_xor(v1, 1);
+                    long i = arow0 + 2 * (lane >> 4) + p;
+                    long j = acol;
+                    bool write = (lane & 1) == 0 && i < n && j < n && i >= j;
+                    if (write) {
+                        float re = (float)(v0 + sv1);
+                        float im = (float)(sv0 - v1);
+                        f2 prev = beta != 0.f ? cb[i * c_row + j] : f2{};
+                        cb[i * c_row + j] = f2{alpha * re + beta * prev.x,
+                                               alpha * im + beta * prev.y};
+                    }
+                }
+            }
+        }
+    }
+}
+
 /* ------------------ specialized correlator cherk (ci8) ------------------ */
 // Per-channel C = X^H.X: C[b][i][j] (i>=j) = alpha*sum_k conj(A[k,i])*A[k,j]
 // + beta*C, with A ci8 k-major: element (k,i) at a + b*a_b + k*lda + i.
@@ -974,6 +1134,26 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                         ((a_k * 2) % 16 == 0) && ((a_b * 2) % 16 == 0);
             bool want_rs = !sel || strcmp(sel, "rs") == 0;
             bool want_wave = sel && strcmp(sel, "wave") == 0;
+            // rs8 (8-wave 128x64 tile) measures ~equal to rs (1.23 vs
+            // 1.25 Gsamp/s at config 3) — opt-in until it wins.
+            const char* sel8 = getenv("BIFROST_CHERK");
+            bool want_rs8 = sel8 && strcmp(sel8, "rs8") == 0;
+            if (al16 && want_rs8 && n % 128 == 0) {
+                long nti = n / 64;           // 64-wide col blocks
+                long nI = n / 128;
+                long ntiles8 = 0;
+                for (long I = 0; I < nI; ++I)
+                    ntiles8 += (2 * I + 2 < nti ? 2 * I + 2 : nti);
+                long nflat8 = ((ntiles8 * nbatch + 7) / 8) * 8;
+                dim3 grid8(cap_grid(nflat8, 65535), 1);
+                hipLaunchKernelGGL(cherk_ci8_mfma_rs8_kernel, grid8,
+                                   dim3(512), 0, stream, n, k, nbatch,
+                                   (float)alpha, (const signed char*)a, a_k,
+                                   a_b, (float)beta, (f2*)c, c_row, c_b,
+                                   nti, ntiles8);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
             if (al16 && want_rs) {
                 const char* bkenv = getenv("BIFROST_CHERK_BK");
                 // BK=64 at 3 waves/SIMD beats BK=128 at 2 (occupancy wins
