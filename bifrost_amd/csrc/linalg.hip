@@ -153,7 +153,7 @@ __device__ inline void lift_tri(long t, long& bi, long& bj) {
 //   a(i,k) = a_base[b*a_bstride + i*a_n + k*a_k]   (element strides)
 //   f = conj(x)*y if CONJ_FIRST else x*conj(y); plain product for reals.
 template <typename Loader, typename Acc, bool CONJ_FIRST>
-__global__ void herk_generic_kernel(long n, long k, long nbatch,
+__global__ __launch_bounds__(256) void herk_generic_kernel(long n, long k, long nbatch,
                                     double alpha, const void* a, long a_n,
                                     long a_k, long a_b, double beta, void* c,
                                     long c_row, long c_b, long ntiles) {
@@ -969,7 +969,7 @@ __global__ void cherk_ci8_kernel(long n, long k, long nbatch, float alpha,
 /* ------------------------ generic gemm (fallback) ----------------------- */
 // C[b][i][j] = alpha * sum_k a^(i,k) * b^(k,j) + beta * C[b][i][j]
 template <typename LoadA, typename LoadB, typename Acc>
-__global__ void gemm_generic_kernel(long m, long nn, long k, long nbatch,
+__global__ __launch_bounds__(256) void gemm_generic_kernel(long m, long nn, long k, long nbatch,
                                     double alpha, const void* a, long a_i,
                                     long a_k, long a_b, int conj_a,
                                     const void* b, long b_k, long b_j,
@@ -1029,14 +1029,107 @@ __global__ void gemm_generic_kernel(long m, long nn, long k, long nbatch,
 //   W: LoadW at w + b*w_b + i*ldw + k      (k-fastest)
 //   X: LoadX at x + b*x_b + j*ldx + k      (k-fastest)
 // MTILE beams per launch chunk kept in registers; W chunk cached in LDS.
+// 8-element vectorized X loads (the element loaders issue per-byte loads,
+// which dominate otherwise: 16 scalar loads per 8 ci8 elements).
+template <typename Loader>
+struct Load8 {
+    __device__ static void load(const void* p, long idx, f2* out) {
+        for (int e = 0; e < 8; ++e) out[e] = Loader::load(p, idx + e);
+    }
+};
+template <>
+struct Load8<LoadCI8<f2>> {
+    __device__ static void load(const void* p, long idx, f2* out) {
+        unsigned d[4];
+        __builtin_memcpy(d, (const signed char*)p + 2 * idx, 16);
+        for (int i = 0; i < 4; ++i) {
+            out[2 * i].x = (float)(signed char)(d[i] & 0xFF);
+            out[2 * i].y = (float)(signed char)((d[i] >> 8) & 0xFF);
+            out[2 * i + 1].x = (float)(signed char)((d[i] >> 16) & 0xFF);
+            out[2 * i + 1].y = (float)(signed char)(d[i] >> 24);
+        }
+    }
+};
+template <>
+struct Load8<LoadCI4<f2>> {
+    __device__ static void load(const void* p, long idx, f2* out) {
+        unsigned d[2];
+        __builtin_memcpy(d, (const signed char*)p + idx, 8);
+        for (int i = 0; i < 8; ++i) {
+            unsigned b = (d[i >> 2] >> (8 * (i & 3))) & 0xFF;
+            out[i].x = (float)((signed char)b >> 4);
+            out[i].y = (float)((signed char)(b << 4) >> 4);
+        }
+    }
+};
+
+// Fast beamform kernel: full 16-beam chunk, k%64==0, nn even — no edge
+// paths at all, so the register allocator keeps the 16x2 accumulators and
+// the vector-load buffers in registers (the mixed-path kernel spills).
 template <typename LoadW, typename LoadX, int MTILE>
-__global__ void beamform_kernel(long mm, long nn, long k, long nbatch,
+__global__ __launch_bounds__(256) void beamform_fast_kernel(
+    long nn, long k, long nbatch, float alpha, const void* w, long ldw,
+    long w_b, const void* x, long ldx, long x_b, float beta,
+    f2* __restrict__ c, long c_row, long c_b, long i0) {
+    __shared__ f2 sw[MTILE][64 + 1];
+    int lane = threadIdx.x;
+    for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
+        long woff = batch * w_b;
+        long xoff = batch * x_b;
+        f2* cb = c + batch * c_b;
+        for (long j0 = (long)blockIdx.x * blockDim.x * 2; j0 < nn;
+             j0 += (long)gridDim.x * blockDim.x * 2) {
+            long j = j0 + 2 * lane;
+            bool ok = j + 1 < nn;
+            f2 acc[MTILE][2];
+            for (int m = 0; m < MTILE; ++m) acc[m][0] = acc[m][1] = f2{};
+            for (long k0 = 0; k0 < k; k0 += 64) {
+                for (int idx = lane; idx < MTILE * 64; idx += 256) {
+                    int m = idx >> 6, kk = idx & 63;
+                    sw[m][kk] = LoadW::load(w, woff + (i0 + m) * ldw + k0 + kk);
+                }
+                __syncthreads();
+                if (ok) {
+                    for (int kg = 0; kg < 64; kg += 8) {
+                        f2 x0[8], x1[8];
+                        Load8<LoadX>::load(x, xoff + j * ldx + k0 + kg, x0);
+                        Load8<LoadX>::load(x, xoff + (j + 1) * ldx + k0 + kg, x1);
+                        for (int kk = 0; kk < 8; ++kk) {
+                            for (int m = 0; m < MTILE; ++m) {
+                                f2 wv = sw[m][kg + kk];
+                                cmac<false, false>(acc[m][0], wv, x0[kk]);
+                                cmac<false, false>(acc[m][1], wv, x1[kk]);
+                            }
+                        }
+                    }
+                }
+                __syncthreads();
+            }
+            if (ok) {
+                for (int m = 0; m < MTILE; ++m) {
+                    for (int e = 0; e < 2; ++e) {
+                        long jj = j + e;
+                        f2 prev = beta != 0.f ? cb[(i0 + m) * c_row + jj] : f2{};
+                        cb[(i0 + m) * c_row + jj] =
+                            f2{alpha * acc[m][e].x + beta * prev.x,
+                               alpha * acc[m][e].y + beta * prev.y};
+                    }
+                }
+            }
+        }
+    }
+}
+
+// Each thread owns TWO consecutive time samples so every W read from LDS
+// feeds two complex MACs (the kernel is otherwise LDS-read-bound).
+template <typename LoadW, typename LoadX, int MTILE>
+__global__ __launch_bounds__(256) void beamform_kernel(long mm, long nn, long k, long nbatch,
                                 float alpha, const void* w, long ldw, long w_b,
                                 const void* x, long ldx, long x_b, float beta,
                                 f2* __restrict__ c, long c_row, long c_b,
                                 long i0) {
     __shared__ f2 sw[MTILE][64 + 1];
-    int lane = threadIdx.x;  // 256 threads; each owns consecutive j
+    int lane = threadIdx.x;  // 256 threads x 2 time samples each
     long mlim = min((long)MTILE, mm - i0);
     for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
         const void* wb = w;
@@ -1044,35 +1137,61 @@ __global__ void beamform_kernel(long mm, long nn, long k, long nbatch,
         long woff = batch * w_b;
         long xoff = batch * x_b;
         f2* cb = c + batch * c_b;
-        for (long j0 = (long)blockIdx.x * blockDim.x; j0 < nn;
-             j0 += (long)gridDim.x * blockDim.x) {
-            long j = j0 + lane;
-            f2 acc[MTILE];
-            for (int m = 0; m < MTILE; ++m) acc[m] = f2{};
+        for (long j0 = (long)blockIdx.x * blockDim.x * 2; j0 < nn;
+             j0 += (long)gridDim.x * blockDim.x * 2) {
+            long j = j0 + 2 * lane;
+            bool ok0 = j < nn, ok1 = j + 1 < nn;
+            f2 acc[MTILE][2];
+            for (int m = 0; m < MTILE; ++m) acc[m][0] = acc[m][1] = f2{};
             for (long k0 = 0; k0 < k; k0 += 64) {
                 int klim = (int)min((long)64, k - k0);
-                // stage W[i0..i0+mlim)[k0..k0+klim) into LDS
                 for (int idx = lane; idx < (int)mlim * klim; idx += 256) {
                     int m = idx / klim, kk = idx % klim;
                     sw[m][kk] = LoadW::load(wb, woff + (i0 + m) * ldw + k0 + kk);
                 }
                 __syncthreads();
-                if (j < nn) {
+                if (ok0 && ok1 && klim == 64 && mlim == MTILE) {
+                    // main path: 8-element vector loads of both X rows and
+                    // a fully unrolled beam loop (a runtime m<mlim guard
+                    // inside the loop forces per-iteration branches and
+                    // kills the pipeline — hoisted here).
+                    for (int kg = 0; kg < 64; kg += 8) {
+                        f2 x0[8], x1[8];
+                        Load8<LoadX>::load(xb, xoff + j * ldx + k0 + kg, x0);
+                        Load8<LoadX>::load(xb, xoff + (j + 1) * ldx + k0 + kg,
+                                           x1);
+                        for (int kk = 0; kk < 8; ++kk) {
+                            for (int m = 0; m < MTILE; ++m) {
+                                f2 wv = sw[m][kg + kk];
+                                cmac<false, false>(acc[m][0], wv, x0[kk]);
+                                cmac<false, false>(acc[m][1], wv, x1[kk]);
+                            }
+                        }
+                    }
+                } else if (ok0) {
                     for (int kk = 0; kk < klim; ++kk) {
-                        f2 xv = LoadX::load(xb, xoff + j * ldx + k0 + kk);
+                        f2 x0 = LoadX::load(xb, xoff + j * ldx + k0 + kk);
+                        f2 x1 = ok1 ? LoadX::load(xb, xoff + (j + 1) * ldx + k0 + kk)
+                                    : f2{};
                         for (int m = 0; m < MTILE; ++m) {
-                            if (m < mlim) cmac<false, false>(acc[m], sw[m][kk], xv);
+                            if (m < mlim) {
+                                f2 wv = sw[m][kk];
+                                cmac<false, false>(acc[m][0], wv, x0);
+                                cmac<false, false>(acc[m][1], wv, x1);
+                            }
                         }
                     }
                 }
                 __syncthreads();
             }
-            if (j < nn) {
-                for (int m = 0; m < (int)mlim; ++m) {
-                    f2 prev = beta != 0.f ? cb[(i0 + m) * c_row + j] : f2{};
-                    cb[(i0 + m) * c_row + j] =
-                        f2{alpha * acc[m].x + beta * prev.x,
-                           alpha * acc[m].y + beta * prev.y};
+            for (int m = 0; m < (int)mlim; ++m) {
+                for (int e = 0; e < 2; ++e) {
+                    if (e == 0 ? !ok0 : !ok1) continue;
+                    long jj = j + e;
+                    f2 prev = beta != 0.f ? cb[(i0 + m) * c_row + jj] : f2{};
+                    cb[(i0 + m) * c_row + jj] =
+                        f2{alpha * acc[m][e].x + beta * prev.x,
+                           alpha * acc[m][e].y + beta * prev.y};
                 }
             }
         }
@@ -1267,8 +1386,27 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
         (a_type == BF_DTYPE_CI16 || a_type == BF_DTYPE_CF32) &&
         c_type == BF_DTYPE_CF32 && a_k == 1 && b_k == 1 && !conj_a &&
         !conj_b && m <= 1024) {
-        dim3 grid(cap_grid((nn + 255) / 256, 4096), cap_grid(nbatch, 65535));
+        dim3 grid(cap_grid((nn + 511) / 512, 4096), cap_grid(nbatch, 65535));
+        bool fast_ok = (k % 64 == 0) && (nn % 2 == 0) && nn >= 2;
         for (long i0 = 0; i0 < m; i0 += 16) {
+            if (fast_ok && i0 + 16 <= m) {
+#define BEAMF_CASE(LW, LX)                                                    \
+    hipLaunchKernelGGL((beamform_fast_kernel<LW<f2>, LX<f2>, 16>), grid,      \
+                       dim3(256), 0, stream, nn, k, nbatch, (float)alpha,     \
+                       a, a_i, a_b, b, b_j, b_b, (float)beta, (f2*)c, c_row,  \
+                       c_b, i0)
+                if (a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI8)
+                    BEAMF_CASE(LoadCF32, LoadCI8);
+                else if (a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI4)
+                    BEAMF_CASE(LoadCF32, LoadCI4);
+                else if (a_type == BF_DTYPE_CI16 && b_type == BF_DTYPE_CI8)
+                    BEAMF_CASE(LoadCI16, LoadCI8);
+                else
+                    BEAMF_CASE(LoadCI16, LoadCI4);
+#undef BEAMF_CASE
+                BF_CHECK_HIP(hipGetLastError());
+                continue;
+            }
 #define BEAM_CASE(LW, LX)                                                     \
     hipLaunchKernelGGL((beamform_kernel<LW<f2>, LX<f2>, 16>), grid,           \
                        dim3(256), 0, stream, m, nn, k, nbatch, (float)alpha,  \

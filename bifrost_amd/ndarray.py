@@ -199,7 +199,10 @@ class ndarray(np.ndarray):
 
     def __del__(self):
         if hasattr(self, "bf") and self.bf.ownbuffer:
-            raw_free(self.bf.ownbuffer, self.bf.space)
+            try:
+                raw_free(self.bf.ownbuffer, self.bf.space)
+            except (AttributeError, TypeError):
+                pass  # interpreter shutdown: module globals already torn down
 
     def as_BFarray(self):
         a = _bf.BFarray()
