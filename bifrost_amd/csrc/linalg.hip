@@ -549,14 +549,33 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
         __syncthreads();
         int buf = 0;
         for (int s = 0; s < nslab; ++s) {
-            // write slab s+1 into the other buffer (its loads were issued
-            // one compute-phase ago), then issue loads for slab s+2 so
-            // they fly across the compute and the bare barrier.
-            if (s + 1 < nslab) {
-                write_slab(buf ^ 1);
-                if (s + 2 < nslab) load_slab();
+            // Issue this slab's fragment reads BEFORE the next slab's
+            // ds_writes: the wave's DS queue is in-order, so writes issued
+            // first would delay every MFMA behind the write drain.  The
+            // fragments live in registers across the staging issue.
+            {
+                const signed char* bI = &lds[buf][0][0][0];
+                const signed char* bJ = &lds[buf][1][0][0];
+                v4i fa[4], fb[4];
+                if (!skip_all) {
+                    for (int ta = 0; ta < 4; ++ta) {
+                        fa[ta] = frag(bI, 4 * wr + ta);
+                        fb[ta] = frag(bJ, 4 * wc + ta);
+                    }
+                }
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+                if (!skip_all) {
+                    for (int ta = 0; ta < 4; ++ta)
+                        for (int tb = 0; tb < 4; ++tb)
+                            acc[ta][tb] =
+                                __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                                    fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+                }
             }
-            for (int h = 0; h < NHALF; ++h) compute(buf, h);
+            for (int h = 1; h < NHALF; ++h) compute(buf, h);
             __syncthreads();
             buf ^= 1;
         }
