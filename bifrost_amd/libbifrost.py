@@ -182,6 +182,12 @@ _proto("bfLinAlgMatMul", BFstatus, _bf.BFlinalg, ctypes.c_double, _PA, _PA,
 _proto("bfTranspose", BFstatus, _PA, _PA, c_int_p)
 _proto("bfUnpack", BFstatus, _PA, _PA, BFbool)
 _proto("bfQuantize", BFstatus, _PA, _PA, ctypes.c_double)
+_proto("bfMap", BFstatus, ctypes.c_int, ctypes.POINTER(ctypes.c_long),
+       ctypes.POINTER(ctypes.c_char_p), ctypes.c_int,
+       ctypes.POINTER(ctypes.POINTER(BFarray)),
+       ctypes.POINTER(ctypes.c_char_p), ctypes.c_char_p, ctypes.c_char_p,
+       ctypes.c_char_p, c_int_p, c_int_p)
+_proto("bfMapClearCache", BFstatus)
 
 # proclog / affinity
 _proto("bfProcLogCreate", BFstatus, ctypes.POINTER(_bf.BFproclog),

@@ -257,9 +257,24 @@ class ndarray(np.ndarray):
                 a = asarray(a)
             a.bf.dtype = dtype_bf
             return a
-        raise NotImplementedError(
-            "astype on device arrays requires the bfMap JIT engine "
-            "(planned; see DESIGN.md)")
+        # device arrays: JIT an elementwise conversion through bfMap.
+        # The package attribute 'map' is rebound to the function by
+        # __init__, so fetch the real module from sys.modules.
+        import importlib
+        bf_map = importlib.import_module("bifrost_amd.map")
+        a = ndarray(shape=self.shape, dtype=dtype_bf, space=self.bf.space)
+        if dtype_bf.is_complex:
+            if self.bf.dtype.is_complex:
+                func = "a.real = b.real; a.imag = b.imag"
+            else:
+                func = "a.real = b; a.imag = 0"
+        else:
+            if self.bf.dtype.is_complex:
+                func = "a = b.real"
+            else:
+                func = "a = b"
+        bf_map.map(func, {"a": a, "b": self})
+        return a
 
     def _system_accessible_copy(self):
         if space_accessible(self.bf.space, ["system"]):
