@@ -1,9 +1,11 @@
 """Block library (hot-path subset per SURVEY.md §2: copy, transpose,
-unpack, quantize, correlate, accumulate)."""
+unpack, quantize, correlate, accumulate, fft, detect)."""
 
 from bifrost_amd.blocks.accumulate import AccumulateBlock, accumulate  # noqa: F401
 from bifrost_amd.blocks.copy import CopyBlock, copy  # noqa: F401
 from bifrost_amd.blocks.correlate import CorrelateBlock, correlate  # noqa: F401
+from bifrost_amd.blocks.detect import DetectBlock, detect  # noqa: F401
+from bifrost_amd.blocks.fft import FftBlock, fft  # noqa: F401
 from bifrost_amd.blocks.quantize import QuantizeBlock, quantize  # noqa: F401
 from bifrost_amd.blocks.transpose import TransposeBlock, transpose  # noqa: F401
 from bifrost_amd.blocks.unpack import UnpackBlock, unpack  # noqa: F401

@@ -119,7 +119,7 @@ _bf.BFspan_info = BFspan_info
 
 # Opaque handles
 for _h in ("BFlinalg", "BFring", "BFsequence", "BFrsequence", "BFwsequence",
-           "BFspan", "BFrspan", "BFwspan", "BFproclog"):
+           "BFspan", "BFrspan", "BFwspan", "BFproclog", "BFfft"):
     setattr(_bf, _h, type(_h, (ctypes.c_void_p,), {}))
 
 _lib = ctypes.CDLL(_LIB_PATH, mode=ctypes.RTLD_GLOBAL)
@@ -188,6 +188,12 @@ _proto("bfMap", BFstatus, ctypes.c_int, ctypes.POINTER(ctypes.c_long),
        ctypes.POINTER(ctypes.c_char_p), ctypes.c_char_p, ctypes.c_char_p,
        ctypes.c_char_p, c_int_p, c_int_p)
 _proto("bfMapClearCache", BFstatus)
+_proto("bfFftCreate", BFstatus, ctypes.POINTER(_bf.BFfft))
+_proto("bfFftDestroy", BFstatus, _bf.BFfft)
+_proto("bfFftInit", BFstatus, _bf.BFfft, _PA, _PA, ctypes.c_int, c_int_p,
+       BFbool, ctypes.POINTER(ctypes.c_size_t))
+_proto("bfFftExecute", BFstatus, _bf.BFfft, _PA, _PA, BFbool,
+       ctypes.c_void_p, ctypes.c_size_t)
 
 # proclog / affinity
 _proto("bfProcLogCreate", BFstatus, ctypes.POINTER(_bf.BFproclog),
