@@ -22,6 +22,7 @@ from bifrost_amd.transpose import transpose  # noqa: F401
 from bifrost_amd.unpack import unpack  # noqa: F401
 from bifrost_amd.map import clear_map_cache, map  # noqa: F401
 from bifrost_amd.fft import Fft  # noqa: F401
+from bifrost_amd.reduce import reduce  # noqa: F401
 
 # Make `import bifrost_amd as bf; bf.DataType.ci8` work like the reference's
 # module attribute access (bifrost.DataType is a module there).

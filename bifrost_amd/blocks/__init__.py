@@ -7,5 +7,6 @@ from bifrost_amd.blocks.correlate import CorrelateBlock, correlate  # noqa: F401
 from bifrost_amd.blocks.detect import DetectBlock, detect  # noqa: F401
 from bifrost_amd.blocks.fft import FftBlock, fft  # noqa: F401
 from bifrost_amd.blocks.quantize import QuantizeBlock, quantize  # noqa: F401
+from bifrost_amd.blocks.reduce import ReduceBlock, reduce  # noqa: F401
 from bifrost_amd.blocks.transpose import TransposeBlock, transpose  # noqa: F401
 from bifrost_amd.blocks.unpack import UnpackBlock, unpack  # noqa: F401

@@ -188,6 +188,7 @@ _proto("bfMap", BFstatus, ctypes.c_int, ctypes.POINTER(ctypes.c_long),
        ctypes.POINTER(ctypes.c_char_p), ctypes.c_char_p, ctypes.c_char_p,
        ctypes.c_char_p, c_int_p, c_int_p)
 _proto("bfMapClearCache", BFstatus)
+_proto("bfReduce", BFstatus, _PA, _PA, ctypes.c_int)
 _proto("bfFftCreate", BFstatus, ctypes.POINTER(_bf.BFfft))
 _proto("bfFftDestroy", BFstatus, _bf.BFfft)
 _proto("bfFftInit", BFstatus, _bf.BFfft, _PA, _PA, ctypes.c_int, c_int_p,
