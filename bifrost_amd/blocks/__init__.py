@@ -2,7 +2,17 @@
 unpack, quantize, correlate, accumulate, fft, detect)."""
 
 from bifrost_amd.blocks.accumulate import AccumulateBlock, accumulate  # noqa: F401
+from bifrost_amd.blocks.binary_io import (BinaryFileReadBlock,  # noqa: F401
+                                          BinaryFileWriteBlock,
+                                          binary_read, binary_write)
 from bifrost_amd.blocks.copy import CopyBlock, copy  # noqa: F401
+from bifrost_amd.blocks.guppi_raw import GuppiRawSourceBlock, read_guppi_raw  # noqa: F401
+from bifrost_amd.blocks.serialize import (DeserializeBlock,  # noqa: F401
+                                          SerializeBlock, deserialize,
+                                          serialize)
+from bifrost_amd.blocks.sigproc import (SigprocSourceBlock,  # noqa: F401
+                                        SigprocSinkBlock, read_sigproc,
+                                        write_sigproc)
 from bifrost_amd.blocks.correlate import CorrelateBlock, correlate  # noqa: F401
 from bifrost_amd.blocks.detect import DetectBlock, detect  # noqa: F401
 from bifrost_amd.blocks.fft import FftBlock, fft  # noqa: F401
