@@ -447,9 +447,19 @@ void cherk_ci8_mfma_kernel(long n, long k, long nbatch, float alpha,
 #ifndef CHERK_RS_W1
 #define CHERK_RS_W1 4
 #endif
-template <int NHALF>  // K-slab = 64*NHALF (barriers amortize with NHALF)
+// SCHED selects the slab schedule (same-box A/B via BIFROST_CHERK_SCHED;
+// sweep results in profiles/round1_cherk.md — 2 is the default):
+//   0 = original: reads -> staging writes -> loads -> MFMA burst
+//   1 = x2-unrolled static lds[0]/lds[1], same order as 0
+//   2 = x2-unrolled static, one-row-pipelined MFMA burst BEFORE writes
+//   3 = schedule 2 at 3 waves/SIMD (more VGPRs, no scratch spill)
+//   4 = schedule 3 + double-buffered stg: loads for slab s+2 issue BEFORE
+//       the ds_writes of slab s+1, so the writes wait vmcnt(4) (the old
+//       loads only) and the new loads get a slab of extra latency slack
+template <int NHALF, int SCHED = 2>
 __global__ __launch_bounds__(256)
-__attribute__((amdgpu_waves_per_eu(NHALF == 1 ? CHERK_RS_W1 : 2)))
+__attribute__((amdgpu_waves_per_eu(
+    NHALF != 1 ? 2 : (SCHED >= 3 ? 3 : CHERK_RS_W1))))
 void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
                               const signed char* __restrict__ a, long lda,
                               long a_b, float beta, f2* __restrict__ c,
@@ -547,6 +557,106 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
         write_slab(0);
         if (nslab > 1) load_slab();
         __syncthreads();
+        if (SCHED >= 1 && NHALF == 1) {
+            v4i stg_b[4];  // second staging set (SCHED==4 only)
+            auto load_into = [&](v4i* dst) {
+                const v4i* pv = (const v4i*)__builtin_assume_aligned(
+                    load_next, 16);
+                load_next += slab_step;
+                dst[0] = pv[0];
+                dst[1] = pv[1];
+                dst[2] = pv[2];
+                dst[3] = pv[3];
+            };
+            auto write_from = [&](auto& dst, const v4i* src4) {
+                int swz = st_row & 7;
+                signed char* base = &dst[st_strip][st_row][0];
+                *(v4i*)(base + 16 * ((st_cq + 0) ^ swz)) = src4[0];
+                *(v4i*)(base + 16 * ((st_cq + 1) ^ swz)) = src4[1];
+                *(v4i*)(base + 16 * ((st_cq + 2) ^ swz)) = src4[2];
+                *(v4i*)(base + 16 * ((st_cq + 3) ^ swz)) = src4[3];
+            };
+            auto burst = [&](const signed char* bI, const signed char* bJ) {
+                if (SCHED == 1) {
+                    v4i fa[4], fb[4];
+                    for (int ta = 0; ta < 4; ++ta) {
+                        fa[ta] = frag(bI, 4 * wr + ta);
+                        fb[ta] = frag(bJ, 4 * wc + ta);
+                    }
+                    for (int ta = 0; ta < 4; ++ta)
+                        for (int tb = 0; tb < 4; ++tb)
+                            acc[ta][tb] =
+                                __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                                    fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+                } else {  // one-row-pipelined burst
+                    v4i fb[4];
+                    for (int tb = 0; tb < 4; ++tb)
+                        fb[tb] = frag(bJ, 4 * wc + tb);
+                    v4i fa_cur = frag(bI, 4 * wr);
+#pragma unroll
+                    for (int ta = 0; ta < 4; ++ta) {
+                        v4i fa_nxt;
+                        if (ta < 3) fa_nxt = frag(bI, 4 * wr + ta + 1);
+                        for (int tb = 0; tb < 4; ++tb)
+                            acc[ta][tb] =
+                                __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                                    fa_cur, fb[tb], acc[ta][tb], 0, 0, 0);
+                        fa_cur = fa_nxt;
+                    }
+                }
+            };
+            auto step = [&](auto& rd, auto& wrbuf, v4i* consume, v4i* fill,
+                            int s) {
+                const signed char* bI = &rd[0][0][0];
+                const signed char* bJ = &rd[1][0][0];
+                if (SCHED == 1) {
+                    // original order: reads -> writes -> loads -> MFMAs
+                    v4i fa[4], fb[4];
+                    if (!skip_all) {
+                        for (int ta = 0; ta < 4; ++ta) {
+                            fa[ta] = frag(bI, 4 * wr + ta);
+                            fb[ta] = frag(bJ, 4 * wc + ta);
+                        }
+                    }
+                    if (s + 1 < nslab) {
+                        write_from(wrbuf, consume);
+                        if (s + 2 < nslab) load_into(fill);
+                    }
+                    if (!skip_all) {
+                        for (int ta = 0; ta < 4; ++ta)
+                            for (int tb = 0; tb < 4; ++tb)
+                                acc[ta][tb] =
+                                    __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                                        fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+                    }
+                } else if (SCHED == 4) {
+                    // burst -> loads (into the other stg set) -> writes:
+                    // the writes wait only the OLD loads (vmcnt(4))
+                    if (!skip_all) burst(bI, bJ);
+                    if (s + 1 < nslab) {
+                        if (s + 2 < nslab) load_into(fill);
+                        write_from(wrbuf, consume);
+                    }
+                } else {
+                    // SCHED 2/3: burst -> writes -> loads (one stg set)
+                    if (!skip_all) burst(bI, bJ);
+                    if (s + 1 < nslab) {
+                        write_from(wrbuf, consume);
+                        if (s + 2 < nslab) load_into(fill);
+                    }
+                }
+                __syncthreads();
+            };
+            int s = 0;
+            while (s < nslab) {
+                step(lds[0], lds[1], stg, SCHED == 4 ? stg_b : stg, s);
+                ++s;
+                if (s >= nslab) break;
+                step(lds[1], lds[0], SCHED == 4 ? stg_b : stg,
+                     stg, s);
+                ++s;
+            }
+        } else {
         int buf = 0;
         for (int s = 0; s < nslab; ++s) {
             // Issue this slab's fragment reads BEFORE the next slab's
@@ -578,6 +688,7 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
             for (int h = 1; h < NHALF; ++h) compute(buf, h);
             __syncthreads();
             buf ^= 1;
+        }
         }
         for (int ta = 0; ta < 4; ++ta) {
             for (int tb = 0; tb < 4; ++tb) {
@@ -1297,18 +1408,27 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 // BK=64 at 3 waves/SIMD beats BK=128 at 2 (occupancy wins
                 // over barrier amortization; measured 1.19 vs 1.11 Gsamp/s)
                 int nhalf = bkenv ? atoi(bkenv) / 64 : 1;
+                const char* schenv = getenv("BIFROST_CHERK_SCHED");
+                int sched = schenv ? atoi(schenv) : 2;
+                auto launch_rs1 = [&](auto kern) {
+                    hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, n,
+                                       k, nbatch, (float)alpha,
+                                       (const signed char*)a, a_k, a_b,
+                                       (float)beta, (f2*)c, c_row, c_b,
+                                       ntiles);
+                };
                 if (nhalf >= 2 && k % 128 == 0)
-                    hipLaunchKernelGGL(cherk_ci8_mfma_rs_kernel<2>, grid,
-                                       dim3(256), 0, stream, n, k, nbatch,
-                                       (float)alpha, (const signed char*)a,
-                                       a_k, a_b, (float)beta, (f2*)c, c_row,
-                                       c_b, ntiles);
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<2, 0>);
+                else if (sched == 0)
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 0>);
+                else if (sched == 1)
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 1>);
+                else if (sched == 3)
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 3>);
+                else if (sched == 4)
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 4>);
                 else
-                    hipLaunchKernelGGL(cherk_ci8_mfma_rs_kernel<1>, grid,
-                                       dim3(256), 0, stream, n, k, nbatch,
-                                       (float)alpha, (const signed char*)a,
-                                       a_k, a_b, (float)beta, (f2*)c, c_row,
-                                       c_b, ntiles);
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 2>);
                 BF_CHECK_HIP(hipGetLastError());
                 return BF_STATUS_SUCCESS;
             }
