@@ -158,17 +158,128 @@ class TestC2R:
                                    rtol=1e-3, atol=1e-2)
 
 
-class TestErrors:
-    def test_nonconsecutive_axes(self):
-        xg = bf.zeros((8, 8, 8), dtype="cf32", space="cuda")
-        yg = bf.zeros((8, 8, 8), dtype="cf32", space="cuda")
-        f = bf.Fft()
-        with pytest.raises(RuntimeError):
-            f.init(xg, yg, axes=[0, 2])
+class TestNonConsecutiveAxes:
+    """Reference test_fft.py dims02/dims03/dims13/dims023... cases: gap
+    dims fold into the plan embeds and host/batch loops."""
 
-    def test_fftshift_unsupported(self):
-        xg = bf.zeros((16,), dtype="cf32", space="cuda")
-        yg = bf.zeros((16,), dtype="cf32", space="cuda")
+    @pytest.mark.parametrize("axes", [[0], [1], [2]])
+    def test_1d_in_3d(self, axes):
+        run_c2c((12, 10, 14), axes, seed=20)
+
+    @pytest.mark.parametrize("axes", [[0, 1], [0, 2], [1, 2]])
+    def test_2d_in_3d(self, axes):
+        run_c2c((12, 10, 14), axes, seed=21)
+
+    @pytest.mark.parametrize("axes", [[0, 1], [0, 2], [0, 3], [1, 2],
+                                      [1, 3], [2, 3]])
+    def test_2d_in_4d(self, axes):
+        run_c2c((6, 8, 10, 12), axes, seed=22)
+
+    @pytest.mark.parametrize("axes", [[0, 1, 2], [0, 1, 3], [0, 2, 3],
+                                      [1, 2, 3]])
+    def test_3d_in_4d(self, axes):
+        run_c2c((6, 8, 10, 12), axes, seed=23)
+
+    @pytest.mark.parametrize("axes", [[0, 2], [1, 3]])
+    def test_inverse_gap(self, axes):
+        run_c2c((6, 8, 10, 12), axes, inverse=True, seed=24)
+
+
+class TestFftshift:
+    """apply_fftshift (c2c): forward output is fftshifted; inverse input
+    is ifftshifted first (reference test_fft.py run_test_c2c_impl)."""
+
+    def _run(self, shape, axes, inverse):
+        x = _rand_c(shape, 30)
+        xg = bf.asarray(x, space="cuda")
+        yg = bf.zeros(shape, dtype="cf32", space="cuda")
+        f = bf.Fft()
+        f.init(xg, yg, axes=axes, apply_fftshift=True)
+        f.execute(xg, yg, inverse=inverse)
+        if inverse:
+            norm = np.prod([shape[a] for a in axes])
+            want = np.fft.ifftn(np.fft.ifftshift(x, axes=axes),
+                                axes=axes) * norm
+        else:
+            want = np.fft.fftshift(np.fft.fftn(x, axes=axes), axes=axes)
+        got = np.asarray(yg.copy("system"))
+        np.testing.assert_allclose(got, want.astype(np.complex64),
+                                   rtol=RTOL, atol=ATOL * np.abs(want).max())
+
+    def test_forward_1d(self):
+        self._run((64,), [0], False)
+
+    def test_forward_1d_odd(self):
+        self._run((33,), [0], False)
+
+    def test_inverse_1d(self):
+        self._run((64,), [0], True)
+
+    def test_inverse_1d_odd(self):
+        self._run((31,), [0], True)
+
+    def test_forward_2d(self):
+        self._run((16, 24), [0, 1], False)
+
+    def test_inverse_2d(self):
+        self._run((16, 24), [0, 1], True)
+
+    def test_forward_batched(self):
+        self._run((8, 32), [1], False)
+
+    def test_forward_gap_axes(self):
+        self._run((8, 6, 10), [0, 2], False)
+
+
+class TestIntegerR2C:
+    """i8/i16 real input converts to f32 scaled by 1/2^(nbit-1)
+    (reference fft_kernels.cu:178-191)."""
+
+    @pytest.mark.parametrize("dtype,scale", [(np.int8, 128.0),
+                                             (np.int16, 32768.0)])
+    def test_1d(self, dtype, scale):
+        rng = np.random.RandomState(40)
+        x = rng.randint(-100, 100, size=(256,)).astype(dtype)
+        xg = bf.asarray(x, space="cuda")
+        yg = bf.zeros((129,), dtype="cf32", space="cuda")
+        f = bf.Fft()
+        f.init(xg, yg, axes=[0])
+        f.execute(xg, yg)
+        want = np.fft.rfft(x.astype(np.float32) / scale)
+        np.testing.assert_allclose(np.asarray(yg.copy("system")), want,
+                                   rtol=1e-3, atol=1e-4 * np.abs(want).max())
+
+    def test_i8_batched(self):
+        rng = np.random.RandomState(41)
+        x = rng.randint(-100, 100, size=(8, 128)).astype(np.int8)
+        xg = bf.asarray(x, space="cuda")
+        yg = bf.zeros((8, 65), dtype="cf32", space="cuda")
+        f = bf.Fft()
+        f.init(xg, yg, axes=[1])
+        f.execute(xg, yg)
+        want = np.fft.rfft(x.astype(np.float32) / 128.0, axis=1)
+        np.testing.assert_allclose(np.asarray(yg.copy("system")), want,
+                                   rtol=1e-3, atol=1e-4 * np.abs(want).max())
+
+    def test_i8_misaligned_1d(self):
+        # odd byte offset into the raw buffer (reference misalign sweep)
+        rng = np.random.RandomState(42)
+        raw = rng.randint(-100, 100, size=(257,)).astype(np.int8)
+        g = bf.asarray(raw, space="cuda")
+        xg = g[1:]
+        yg = bf.zeros((129,), dtype="cf32", space="cuda")
+        f = bf.Fft()
+        f.init(xg, yg, axes=[0])
+        f.execute(xg, yg)
+        want = np.fft.rfft(raw[1:].astype(np.float32) / 128.0)
+        np.testing.assert_allclose(np.asarray(yg.copy("system")), want,
+                                   rtol=1e-3, atol=1e-4 * np.abs(want).max())
+
+
+class TestErrors:
+    def test_fftshift_r2c_unsupported(self):
+        xg = bf.zeros((16,), dtype="f32", space="cuda")
+        yg = bf.zeros((9,), dtype="cf32", space="cuda")
         f = bf.Fft()
         with pytest.raises(RuntimeError):
             f.init(xg, yg, axes=[0], apply_fftshift=True)
