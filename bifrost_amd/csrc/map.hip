@@ -1,9 +1,15 @@
 // bifrost_amd: bfMap — the user-function JIT engine,
 // re-built on hipRTC for gfx950.  Behaviour contract: reference
-// src/map.cpp:110-605 semantics for the ELEMENTWISE surface (named arrays
-// + scalars, numpy-style trailing broadcast, .real/.imag access,
-// extra_code); the explicit axis-indexed form `c(i,j) = ...` returns
-// BF_STATUS_UNSUPPORTED this round.
+// src/map.cpp:110-605 semantics:
+//   * elementwise surface: named arrays + scalars, numpy-style trailing
+//     broadcast, .real/.imag access, extra_code;
+//   * explicit axis-indexed form `c(i,j) = a(j,i)` with shape/axis_names:
+//     per-arg accessor structs with embedded constant shapes/strides,
+//     tail-aligned index broadcasting and Python-style single-wrap
+//     negative indices (reference ArrayIndexer.cuh:66-82), the implicit
+//     index vector `_` (IndexVec arithmetic, reference IndexArray.cuh)
+//     and `.shape()`;
+//   * packed sub-byte dtypes (ci4 etc.) are not supported in map.
 //
 // Codegen strategy: computation shape and every argument's (broadcast-
 // aligned) shape/strides are embedded as compile-time constants, so index
@@ -15,6 +21,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hiprtc.h>
 
+#include <cctype>
 #include <cstring>
 #include <list>
 #include <mutex>
@@ -69,8 +76,31 @@ struct Complex {
     __device__ Complex operator*(T s) const {
         return Complex(real * s, imag * s);
     }
+    __device__ Complex& operator+=(const Complex& o) {
+        real += o.real; imag += o.imag; return *this;
+    }
+    __device__ Complex& operator-=(const Complex& o) {
+        real -= o.real; imag -= o.imag; return *this;
+    }
+    __device__ Complex& operator*=(const Complex& o) {
+        T r = real * o.real - imag * o.imag;
+        imag = real * o.imag + imag * o.real;
+        real = r; return *this;
+    }
     __device__ T mag2() const { return real * real + imag * imag; }
+    __device__ Complex conj() const { return Complex(real, -imag); }
+    __device__ Complex& assign(T r, T i) {
+        real = r; imag = i; return *this;
+    }
 };
+template<typename T>
+__device__ Complex<T> operator*(T s, const Complex<T>& c) {
+    return Complex<T>(s * c.real, s * c.imag);
+}
+template<typename T>
+__device__ Complex<T> operator*(int s, const Complex<T>& c) {
+    return Complex<T>((T)s * c.real, (T)s * c.imag);
+}
 template<typename T>
 __device__ Complex<T> conj(const Complex<T>& c) {
     return Complex<T>(c.real, -c.imag);
@@ -78,8 +108,46 @@ __device__ Complex<T> conj(const Complex<T>& c) {
 typedef Complex<signed char> ci8_t;
 typedef Complex<short>       ci16_t;
 typedef Complex<int>         ci32_t;
+typedef Complex<long long>   ci64_t;
 typedef Complex<float>       cf32_t;
 typedef Complex<double>      cf64_t;
+
+template<int N> struct IndexVec {
+    long v[N];
+    __device__ long operator[](int i) const { return v[i]; }
+};
+#define IV_OP(op)                                                        \
+template<int N>                                                          \
+__device__ IndexVec<N> operator op(const IndexVec<N>& a,                 \
+                                   const IndexVec<N>& b) {               \
+    IndexVec<N> r;                                                       \
+    for (int i = 0; i < N; ++i) r.v[i] = a.v[i] op b.v[i];               \
+    return r;                                                            \
+}                                                                        \
+template<int N>                                                          \
+__device__ IndexVec<N> operator op(const IndexVec<N>& a, long b) {       \
+    IndexVec<N> r;                                                       \
+    for (int i = 0; i < N; ++i) r.v[i] = a.v[i] op b;                    \
+    return r;                                                            \
+}
+IV_OP(+)
+IV_OP(-)
+IV_OP(*)
+IV_OP(/)
+IV_OP(%)
+
+template<typename S>
+__device__ inline void iv_put(long* b, int& n, S x) { b[n++] = (long)x; }
+template<int K>
+__device__ inline void iv_put(long* b, int& n, const IndexVec<K>& x) {
+    for (int i = 0; i < K; ++i) b[n++] = x.v[i];
+}
+__device__ inline void iv_puts(long*, int&) {}
+template<typename T0, typename... Ts>
+__device__ inline void iv_puts(long* b, int& n, T0 a, Ts... rest) {
+    iv_put(b, n, a);
+    iv_puts(b, n, rest...);
+}
 )";
 
 std::string dtype_ctype(BFdtype dt) {
@@ -194,14 +262,30 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
     (void)block_shape;
     (void)block_axes;
 
-    // Explicit axis-indexed form `name(i,...)` is not supported this round.
+    // Which args are used in the explicit indexed form `name(...)` /
+    // `name.shape()`?  Those get accessor structs; plain uses get a bound
+    // reference (the elementwise path, constant-folded offsets).
+    bool indexed[16] = {false};
+    bool any_indexed = false;
     for (int a = 0; a < narg; ++a) {
-        std::string pat = std::string(arg_names[a]) + "(";
-        if (std::strstr(func, pat.c_str()))
-            return BF_STATUS_UNSUPPORTED;
+        std::string pat1 = std::string(arg_names[a]) + "(";
+        std::string pat2 = std::string(arg_names[a]) + ".shape";
+        std::string pat3 = std::string(arg_names[a]) + " (";
+        if (std::strstr(func, pat1.c_str()) ||
+            std::strstr(func, pat2.c_str()) ||
+            std::strstr(func, pat3.c_str())) {
+            indexed[a] = true;
+            any_indexed = true;
+        }
     }
-    if (axis_names && ndim > 0) {
-        // axis names only matter for the indexed form; accept and ignore
+    // bare `_` (the implicit index vector)?
+    bool uses_underscore = false;
+    for (const char* p = std::strchr(func, '_'); p;
+         p = std::strchr(p + 1, '_')) {
+        bool lok = (p == func) ||
+                   (!isalnum((unsigned char)p[-1]) && p[-1] != '_');
+        bool rok = !isalnum((unsigned char)p[1]) && p[1] != '_';
+        if (lok && rok) { uses_underscore = true; break; }
     }
 
     // Computation shape: explicit, or broadcast of the args.
@@ -241,35 +325,97 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
     std::ostringstream os;
     os << kPrelude;
     if (extra_code) os << extra_code << "\n";
+    // accessor structs for indexed args: shapes/strides as literals;
+    // tail-aligned index broadcast with single-wrap negative indices
+    // (reference ArrayIndexer.cuh:66-82)
+    for (int a = 0; a < narg; ++a) {
+        if (!indexed[a]) continue;
+        const BFarray* arr = args[a];
+        std::string ct = dtype_ctype(arr->dtype);
+        int andim = arr->ndim;
+        os << "struct Acc_" << arg_names[a] << " {\n";
+        os << "  char* p;\n";
+        os << "  __device__ static long shp(int d) { return ";
+        for (int d = 0; d < andim; ++d)
+            os << "d==" << d << " ? " << arr->shape[d] << "L : ";
+        os << "1L; }\n";
+        os << "  __device__ static long str(int d) { return ";
+        for (int d = 0; d < andim; ++d)
+            os << "d==" << d << " ? " << arr->strides[d] << "L : ";
+        os << "0L; }\n";
+        os << "  __device__ IndexVec<" << andim << "> shape() const {\n";
+        os << "    IndexVec<" << andim << "> r = {{";
+        for (int d = 0; d < andim; ++d)
+            os << (d ? ", " : "") << arr->shape[d] << "L";
+        os << "}}; return r; }\n";
+        os << "  __device__ " << ct << "& ref(const long* ind, int nind) "
+              "const {\n";
+        os << "    long off = 0;\n";
+        os << "    int lead = nind - " << andim << "; "
+              "if (lead < 0) lead = 0;\n";
+        os << "    int nd = nind < " << andim << " ? nind : " << andim
+           << ";\n";
+        os << "    for (int d = 0; d < nd; ++d) {\n";
+        os << "      long i2 = ind[d + lead];\n";
+        os << "      i2 += (i2 < 0) * shp(d);\n";
+        os << "      off += (shp(d) != 1) * i2 * str(d);\n";
+        os << "    }\n";
+        os << "    return *(" << ct << "*)(p + off);\n";
+        os << "  }\n";
+        os << "  template<typename... Is>\n";
+        os << "  __device__ " << ct << "& operator()(Is... is) const {\n";
+        os << "    long buf[16]; int n2 = 0; iv_puts(buf, n2, is...);\n";
+        os << "    return ref(buf, n2);\n";
+        os << "  }\n";
+        os << "};\n";
+    }
     os << "extern \"C\" __global__ void bfmap_kernel(";
     for (int a = 0; a < narg; ++a) {
-        os << (a ? ", " : "") << "char* __restrict__ p" << a;
+        os << (a ? ", " : "") << "char* __restrict__ _bf_p" << a;
     }
     os << ") {\n";
-    os << "  const long N = " << n << "L;\n";
-    os << "  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x;\n"
-          "       i < N; i += (long)gridDim.x * blockDim.x) {\n";
+    os << "  const long _bf_N = " << n << "L;\n";
+    os << "  for (long _bf_i = (long)blockIdx.x * blockDim.x + threadIdx.x;\n"
+          "       _bf_i < _bf_N; _bf_i += (long)gridDim.x * blockDim.x) {\n";
     // decompose i (row-major over cshape) and compute per-arg offsets
-    os << "    long rem = i;\n";
+    os << "    long _bf_rem = _bf_i;\n";
     for (int d = cndim - 1; d >= 0; --d) {
-        os << "    long idx" << d << " = rem % " << cshape[d] << "L; "
-           << "rem /= " << cshape[d] << "L;\n";
+        os << "    long _bf_idx" << d << " = _bf_rem % " << cshape[d]
+           << "L; _bf_rem /= " << cshape[d] << "L;\n";
+    }
+    // named axes (explicit-shape indexed form)
+    if (axis_names) {
+        for (int d = 0; d < cndim; ++d) {
+            if (axis_names[d] && axis_names[d][0])
+                os << "    long " << axis_names[d] << " = _bf_idx" << d
+                   << ";\n";
+        }
+    }
+    if (uses_underscore || any_indexed) {
+        os << "    IndexVec<" << cndim << "> _ = {{";
+        for (int d = 0; d < cndim; ++d) os << (d ? ", " : "") << "_bf_idx" << d;
+        os << "}};\n    (void)_;\n";
     }
     for (int a = 0; a < narg; ++a) {
         const BFarray* arr = args[a];
+        if (indexed[a]) {
+            os << "    Acc_" << arg_names[a] << " " << arg_names[a]
+               << "{_bf_p" << a << "};\n";
+            continue;
+        }
         int off = cndim - arr->ndim;
-        os << "    long off" << a << " = 0";
+        os << "    long _bf_off" << a << " = 0";
         for (int d = 0; d < arr->ndim; ++d) {
             long ext = arr->shape[d];
             long strd = arr->strides[d];
-            if (ext != 1 && strd != 0) {
-                os << " + idx" << (off + d) << " * " << strd << "L";
+            if (ext != 1 && strd != 0 && off + d >= 0) {
+                os << " + _bf_idx" << (off + d) << " * " << strd << "L";
             }
         }
         os << ";\n";
         std::string ct = dtype_ctype(arr->dtype);
         os << "    " << ct << "& " << arg_names[a] << " = *(" << ct
-           << "*)(p" << a << " + off" << a << ");\n";
+           << "*)(_bf_p" << a << " + _bf_off" << a << ");\n";
     }
     os << "    " << func << ";\n";
     os << "  }\n}\n";

@@ -13,12 +13,15 @@ __all__ = ["map", "clear_map_cache"]
 
 def map(func_string, data, axis_names=None, shape=None, func_name=None,
         extra_code=None, block_shape=None, block_axes=None):
-    """Apply `func_string` elementwise to the named arrays in `data`.
+    """Apply `func_string` to the named arrays in `data`.
 
     Examples::
         bf.map("c = a + b", {'c': c, 'a': a, 'b': b})
         bf.map("a = c.real; b = c.imag", {'c': c, 'a': a, 'b': b})
         bf.map("c = a * s", {'c': c, 'a': a, 's': 2.0})
+        bf.map("b(i,j) = a(j,i)", {'a': a, 'b': b},
+               axis_names=('i', 'j'), shape=b.shape)
+        bf.map("b = a(_ - a.shape()/2)", {'a': a, 'b': b})  # fftshift
     """
     narg = len(data)
     names = []

@@ -96,9 +96,116 @@ class TestDeviceAstype:
         np.testing.assert_array_equal(np.asarray(out.copy("system")),
                                       c.real)
 
-    def test_unsupported_indexed_form(self):
-        a = bf.zeros((4, 4), dtype="f32", space="cuda")
+    def test_indexed_astype_roundabout(self):
+        # the indexed form now powers reference-style expressions too
+        a = bf.asarray(np.arange(16, dtype=np.float32).reshape(4, 4),
+                       space="cuda")
         c = bf.zeros((4, 4), dtype="f32", space="cuda")
-        with pytest.raises(RuntimeError):
-            bf.map("c(i,j) = a(j,i)", {"c": c, "a": a},
-                   axis_names=("i", "j"), shape=(4, 4))
+        bf.map("c(i,j) = a(j,i)", {"c": c, "a": a},
+               axis_names=("i", "j"), shape=(4, 4))
+        got = np.asarray(c.copy("system"))
+        np.testing.assert_array_equal(got,
+                                      np.arange(16).reshape(4, 4).T)
+
+
+class TestIndexedForm:
+    """Reference test_map.py explicit-indexing cases."""
+
+    def test_explicit_indexing_transpose(self):
+        # reference test_map.py:209 (b(i,j,k) = a(j,k,i))
+        rng = np.random.RandomState(50)
+        a = rng.randint(0, 65536, size=(15, 26, 37)).astype(np.int32)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.zeros((37, 15, 26), dtype="i32", space="cuda")
+        bf.map("b(i,j,k) = a(j,k,i)", shape=bg.shape,
+               axis_names=("i", "j", "k"), data={"a": ag, "b": bg})
+        np.testing.assert_array_equal(np.asarray(bg.copy("system")),
+                                      a.transpose(2, 0, 1))
+
+    def test_custom_shape_slice(self):
+        # reference test_map.py:220 (b(i,k) = a(i,j,k), scalar j)
+        rng = np.random.RandomState(51)
+        a = rng.randint(0, 65536, size=(15, 26, 37)).astype(np.int32)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.zeros((15, 37), dtype="i32", space="cuda")
+        bf.map("b(i,k) = a(i,j,k)", shape=bg.shape, axis_names=("i", "k"),
+               data={"a": ag, "b": bg, "j": 11})
+        np.testing.assert_array_equal(np.asarray(bg.copy("system")),
+                                      a[:, 11, :])
+
+    def test_shift_fftshift(self):
+        # reference test_map.py:133 (b = a(_-a.shape()/2) == fftshift)
+        rng = np.random.RandomState(52)
+        a = rng.randint(0, 65536, size=(15, 26, 37)).astype(np.int32)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.zeros(a.shape, dtype="i32", space="cuda")
+        bf.map("b = a(_-a.shape()/2)", data={"a": ag, "b": bg})
+        np.testing.assert_array_equal(np.asarray(bg.copy("system")),
+                                      np.fft.fftshift(a))
+
+    def test_polarisation_products(self):
+        # reference test_map.py:186
+        n = 89
+        rng = np.random.RandomState(53)
+        a = (rng.randint(-127, 128, size=(n, 2)) +
+             1j * rng.randint(-127, 128, size=(n, 2))) \
+            .astype(np.complex64)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.zeros((n, 2), dtype="cf32", space="cuda")
+        bf.map("""
+        auto x = a(_,0);
+        auto y = a(_,1);
+        b(_,0).assign(x.mag2(), y.mag2());
+        b(_,1) = x*y.conj();
+        """, shape=bg.shape[:-1], data={"a": ag, "b": bg})
+        got = np.asarray(bg.copy("system"))
+        def mag2(x):
+            return x.real * x.real + x.imag * x.imag
+        gold = np.empty_like(a)
+        gold[:, 0] = mag2(a[:, 0]) + 1j * mag2(a[:, 1])
+        gold[:, 1] = a[:, 0] * a[:, 1].conj()
+        np.testing.assert_array_equal(got, gold)
+
+    @pytest.mark.parametrize("in_dtype", ["ci8", "ci16", "ci32"])
+    @pytest.mark.parametrize("out_kind", ["same", "cf32"])
+    def test_complex_integer_copy(self, in_dtype, out_kind):
+        # reference test_map.py:153
+        n = 797
+        rng = np.random.RandomState(54)
+        a = bf.ndarray(shape=(n,), dtype=in_dtype, space="system")
+        a["re"] = rng.randint(-100, 100, size=n)
+        a["im"] = rng.randint(-100, 100, size=n)
+        gold = a["re"].astype(np.float32) + 1j * a["im"]
+        out_dtype = in_dtype if out_kind == "same" else "cf32"
+        ag = a.copy(space="cuda")
+        bg = bf.ndarray(shape=(n,), dtype=out_dtype, space="cuda")
+        bf.map("b(i) = a(i)", {"a": ag, "b": bg}, shape=ag.shape,
+               axis_names=("i",))
+        b = bg.copy(space="system")
+        if out_kind == "same":
+            got = b["re"].astype(np.float32) + 1j * b["im"]
+        else:
+            got = np.asarray(b)
+        np.testing.assert_array_equal(got, gold)
+
+    def test_simple_funcs_match_reference(self):
+        # reference run_simple_test_funcs: pow/rint, auto tmp, +=
+        x = np.random.RandomState(55).randint(256, size=797)
+        xg = bf.asarray(x, space="cuda")
+        for funcstr, f in [("y = x+1", lambda v: v + 1),
+                           ("y = x*3", lambda v: v * 3),
+                           ("y = rint(pow(x, 2.f))", lambda v: v ** 2),
+                           ("auto tmp = x; y = tmp*tmp", lambda v: v * v),
+                           ("y = x; y += x", lambda v: v + v)]:
+            yg = bf.zeros(x.shape, dtype="i64", space="cuda")
+            bf.map(funcstr, {"x": xg, "y": yg})
+            np.testing.assert_array_equal(
+                np.asarray(yg.copy("system")), f(x), err_msg=funcstr)
+
+    def test_negative_index_wrap(self):
+        a = bf.asarray(np.arange(10, dtype=np.float32), space="cuda")
+        b = bf.zeros((10,), dtype="f32", space="cuda")
+        bf.map("b(i) = a(i - 10)", {"a": a, "b": b}, shape=(10,),
+               axis_names=("i",))
+        np.testing.assert_array_equal(np.asarray(b.copy("system")),
+                                      np.arange(10))
