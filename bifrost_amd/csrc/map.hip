@@ -369,6 +369,12 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
         os << "  }\n";
         os << "};\n";
     }
+    // <name>_type typedefs (the reference exposes each arg's element
+    // type under this name, e.g. Complex<b_type>)
+    for (int a = 0; a < narg; ++a) {
+        os << "typedef " << dtype_ctype(args[a]->dtype) << " "
+           << arg_names[a] << "_type;\n";
+    }
     os << "extern \"C\" __global__ void bfmap_kernel(";
     for (int a = 0; a < narg; ++a) {
         os << (a ? ", " : "") << "char* __restrict__ _bf_p" << a;
