@@ -16,6 +16,11 @@ from bifrost_amd.blocks.sigproc import (SigprocSourceBlock,  # noqa: F401
 from bifrost_amd.blocks.correlate import CorrelateBlock, correlate  # noqa: F401
 from bifrost_amd.blocks.detect import DetectBlock, detect  # noqa: F401
 from bifrost_amd.blocks.fft import FftBlock, fft  # noqa: F401
+from bifrost_amd.blocks.fftshift import FftShiftBlock, fftshift  # noqa: F401
+from bifrost_amd.blocks.print_header import (PrintHeaderBlock,  # noqa: F401
+                                             print_header)
+from bifrost_amd.blocks.reverse import ReverseBlock, reverse  # noqa: F401
+from bifrost_amd.blocks.scrunch import ScrunchBlock, scrunch  # noqa: F401
 from bifrost_amd.blocks.quantize import QuantizeBlock, quantize  # noqa: F401
 from bifrost_amd.blocks.reduce import ReduceBlock, reduce  # noqa: F401
 from bifrost_amd.blocks.transpose import TransposeBlock, transpose  # noqa: F401

@@ -343,6 +343,7 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
         for (int d = 0; d < andim; ++d)
             os << "d==" << d << " ? " << arr->strides[d] << "L : ";
         os << "0L; }\n";
+        os << "  __device__ static long shape(int d) { return shp(d); }\n";
         os << "  __device__ IndexVec<" << andim << "> shape() const {\n";
         os << "    IndexVec<" << andim << "> r = {{";
         for (int d = 0; d < andim; ++d)
