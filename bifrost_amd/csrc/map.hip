@@ -112,6 +112,16 @@ typedef Complex<long long>   ci64_t;
 typedef Complex<float>       cf32_t;
 typedef Complex<double>      cf64_t;
 
+template<typename T, int N> struct Vec {
+    T v[N];
+    typedef T value_type;
+    __device__ Vec() {}
+    template<typename... As>
+    __device__ Vec(As... as) : v{T(as)...} {}
+    __device__ T& operator[](int i) { return v[i]; }
+    __device__ const T& operator[](int i) const { return v[i]; }
+};
+
 template<int N> struct IndexVec {
     long v[N];
     __device__ long operator[](int i) const { return v[i]; }
@@ -154,6 +164,7 @@ std::string dtype_ctype(BFdtype dt) {
     bool cplx = dt & BF_DTYPE_COMPLEX_BIT;
     int nbit = dt & BF_DTYPE_NBIT_BITS;
     int type = dt & BF_DTYPE_TYPE_BITS;
+    int veclen = ((dt & BF_DTYPE_VECTOR_BITS) >> BF_DTYPE_VECTOR_BIT0) + 1;
     std::ostringstream os;
     if (cplx) {
         if (type == BF_DTYPE_FLOAT_TYPE) os << "cf" << nbit << "_t";
@@ -163,6 +174,8 @@ std::string dtype_ctype(BFdtype dt) {
         else if (type == BF_DTYPE_UINT_TYPE) os << "u" << nbit;
         else os << "i" << nbit;
     }
+    if (veclen > 1) return "Vec<" + os.str() + "," +
+                           std::to_string(veclen) + ">";
     return os.str();
 }
 
