@@ -1,0 +1,136 @@
+"""ndarray surface tests (GPU parts), mirroring reference
+test/test_ndarray.py CUDA cases."""
+
+import ctypes
+
+import numpy as np
+import pytest
+
+import bifrost_amd as bf
+from bifrost_amd.DataType import DataType
+
+pytestmark = pytest.mark.gpu
+
+KNOWN_VALS = [[0, 1], [2, 3], [4, 5]]
+KNOWN = np.array(KNOWN_VALS, dtype=np.float32)
+
+
+def _raw_bytes_equal(x, y):
+    xd = ctypes.cast(x.ctypes.data, ctypes.POINTER(ctypes.c_double))
+    yd = ctypes.cast(y.ctypes.data, ctypes.POINTER(ctypes.c_double))
+    np.testing.assert_equal([xd[i] for i in range(x.size)],
+                            [yd[i] for i in range(y.size)])
+
+
+def test_space_copy():
+    c = bf.ndarray(KNOWN_VALS, dtype="f32")
+    c = c.copy(space="cuda").copy(space="cuda_host").copy(space="system")
+    np.testing.assert_equal(np.asarray(c), KNOWN)
+
+
+def test_space_contiguous_copy():
+    a = np.random.rand(2, 3, 4, 5)
+    b = a.transpose(0, 3, 2, 1).copy()
+    c = bf.zeros(a.shape, dtype=a.dtype, space="system")
+    c[...] = a
+    c = c.copy(space="cuda")
+    d = c.transpose(0, 3, 2, 1).copy(space="system")
+    _raw_bytes_equal(d, b)
+
+
+def test_space_slice_copy():
+    a = np.random.rand(2, 3, 4, 5)
+    b = a[:, 1:, :, :].copy()
+    c = bf.zeros(a.shape, dtype=a.dtype, space="system")
+    c[...] = a
+    c = c.copy(space="cuda")
+    d = c[:, 1:, :, :].copy(space="system")
+    _raw_bytes_equal(d, b)
+
+
+def test_space_contiguous_slice_copy():
+    a = np.random.rand(2, 3, 4, 5)
+    b = a.transpose(0, 3, 2, 1)[:, 1:, :, :].copy()
+    c = bf.zeros(a.shape, dtype=a.dtype, space="system")
+    c[...] = a
+    c = c.copy(space="cuda")
+    d = c.transpose(0, 3, 2, 1)[:, 1:, :, :].copy(space="system")
+    _raw_bytes_equal(d, b)
+
+
+def test_str():
+    e = bf.ndarray(KNOWN_VALS, dtype="f32", space="cuda")
+    assert str(e) == str(KNOWN)
+
+
+def test_repr():
+    f = bf.ndarray(KNOWN_VALS, dtype="f32", space="cuda")
+    repr_f = repr(f)[repr(f).find("("):].replace(" ", "")
+    repr_k = repr(KNOWN)[repr(KNOWN).find("("):].replace(" ", "")
+    assert repr_f == repr_k
+
+
+def test_zeros_like():
+    g = bf.ndarray(KNOWN_VALS, dtype="f32", space="cuda")
+    g = bf.zeros_like(g)
+    np.testing.assert_equal(np.asarray(g.copy("system")),
+                            np.zeros_like(KNOWN))
+
+
+def test_getitem():
+    g = bf.asarray(KNOWN, space="cuda")
+    np.testing.assert_equal(np.asarray(g[0].copy("system")), KNOWN[0])
+    np.testing.assert_equal(np.asarray(g[(0,)].copy("system")),
+                            KNOWN[(0,)])
+    assert float(np.asarray(g[1:, 1:].copy("system"))[0, 0]) == KNOWN[1, 1]
+    np.testing.assert_equal(np.asarray(g[:1, 1:].copy("system")),
+                            KNOWN[:1, 1:])
+
+
+def test_setitem():
+    g = bf.zeros_like(KNOWN_VALS, space="cuda")
+    g[...] = KNOWN_VALS
+    np.testing.assert_equal(np.asarray(g.copy("system")), KNOWN)
+    g[:1, 1:] = [[999]]
+    np.testing.assert_equal(np.asarray(g.copy("system")),
+                            np.array([[0, 999], [2, 3], [4, 5]]))
+    g[0] = [99, 88]
+    np.testing.assert_equal(np.asarray(g.copy("system")),
+                            np.array([[99, 88], [2, 3], [4, 5]]))
+    g[:, 1] = [77, 66, 55]
+    np.testing.assert_equal(np.asarray(g.copy("system")),
+                            np.array([[99, 77], [2, 66], [4, 55]]))
+
+
+def test_space_type_conversion():
+    # reference run_type_conversion(space='cuda'): device-side astype
+    for dtype_in in (np.int8, np.int16, np.int32, np.float32):
+        a = np.array(KNOWN_VALS, dtype=dtype_in)
+        c = bf.ndarray(a, space="cuda")
+        for dtype in ("i8", "i16", "i32", "i64", "f64", "ci8", "ci16",
+                      "ci32", "cf32", "cf64"):
+            np_dtype = DataType(dtype).as_numpy_dtype()
+            try:
+                len(np_dtype)
+                b = np.zeros(a.shape, dtype=np_dtype)
+                b["re"] = a
+            except (IndexError, TypeError):
+                b = a.astype(np_dtype)
+            d = c.astype(dtype).copy(space="system")
+            np.testing.assert_equal(b, np.asarray(d), err_msg="%s->%s" %
+                                    (dtype_in, dtype))
+    a = np.array(KNOWN_VALS, dtype=np.float32)
+    a = np.stack([a, a[::-1]], axis=0).view(np.complex64)
+    c = bf.ndarray(a, space="cuda")
+    for dtype in ("ci8", "ci16", "ci32", "cf32", "cf64", "f64"):
+        np_dtype = DataType(dtype).as_numpy_dtype()
+        try:
+            len(np_dtype)
+            b = np.zeros(a.shape, dtype=np_dtype)
+            b["re"] = a.real
+            b["im"] = a.imag
+        except (IndexError, TypeError):
+            b = a.astype(np_dtype)
+        d = c.astype(dtype).copy(space="system")
+        np.testing.assert_equal(b, np.asarray(d), err_msg="cf32->%s" %
+                                dtype)
