@@ -13,6 +13,7 @@ from bifrost_amd import pipeline  # noqa: F401
 from bifrost_amd.pipeline import Pipeline, block_scope, get_default_pipeline  # noqa: F401
 from bifrost_amd.ring2 import Ring  # noqa: F401
 from bifrost_amd import blocks  # noqa: F401
+from bifrost_amd import views  # noqa: F401
 from bifrost_amd.block_chainer import BlockChainer  # noqa: F401
 from bifrost_amd.DataType import DataType  # noqa: F401
 from bifrost_amd.ndarray import (asarray, copy_array, empty, empty_like,  # noqa: F401
