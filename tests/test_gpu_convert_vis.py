@@ -103,3 +103,69 @@ class TestConvertVisibilities:
         want = _full_from_lower(m)
         np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-5)
         assert hdr["_tensor"]["labels"] == MATRIX_LABELS
+
+
+class TestCorrelatePipeline:
+    """Reference test_pipeline.py test_correlate/test_convert_visibilities:
+    CorrelateBlock feeding the conversion chains, on constant-in-time
+    voltages so the integrated result is nreduce * outer product."""
+
+    def _voltages(self, ntime, nchan, nstand, npol):
+        # per reference CorrelateTestInputBlock: station/pol pattern
+        # constant over time and channel
+        i = np.arange(nstand * npol * 2) % 255 - 127
+        x = np.empty((nchan, nstand * npol), dtype=np.complex64)
+        x.real = i[0::2]
+        x.imag = i[1::2]
+        v = np.broadcast_to(x, (ntime, nchan, nstand * npol))
+        raw = np.empty((ntime, nchan, nstand, npol, 2), dtype=np.int8)
+        raw[..., 0] = v.real.reshape(ntime, nchan, nstand, npol)
+        raw[..., 1] = v.imag.reshape(ntime, nchan, nstand, npol)
+        return raw, x
+
+    def test_correlate_then_convert(self):
+        ntime, nchan, nstand, npol = 400, 8, 12, 2
+        nreduce = 100
+        raw, x = self._voltages(ntime, nchan, nstand, npol)
+        data = bf.ndarray(raw.view(bf.DataType.ci8)
+                          .reshape(ntime, nchan, nstand, npol))
+
+        full_out, low_out, storage_rt = [], [], []
+        with bf.Pipeline() as pipe:
+            src = NumpySourceBlock([data], gulp_nframe=100,
+                                   labels=["time", "freq", "station",
+                                           "pol"])
+            dev = bf.blocks.copy(src, space="cuda")
+            vis = bf.blocks.correlate(dev, nreduce)
+            CollectBlock(bf.blocks.copy(vis, space="cuda_host"), low_out)
+            fullm = bf.blocks.convert_visibilities(vis, "matrix")
+            CollectBlock(bf.blocks.copy(fullm, space="cuda_host"),
+                         full_out)
+            stor = bf.blocks.convert_visibilities(vis, "storage")
+            backm = bf.blocks.convert_visibilities(stor, "matrix")
+            CollectBlock(bf.blocks.copy(backm, space="cuda_host"),
+                         storage_rt)
+            pipe.run()
+
+        n = nstand * npol
+        expected = nreduce * x[:, :, None].conj() * x[:, None, :]
+        expected = np.broadcast_to(
+            expected.reshape(1, nchan, nstand, npol, nstand, npol),
+            (ntime // nreduce, nchan, nstand, npol, nstand, npol))
+
+        # lower-only output: compare the lower triangle
+        low = np.concatenate(low_out, axis=0) \
+            .reshape(-1, nchan, n, n)
+        exp_flat = np.asarray(expected).reshape(-1, nchan, n, n)
+        tril = np.tril_indices(n)
+        np.testing.assert_allclose(low[..., tril[0], tril[1]],
+                                   exp_flat[..., tril[0], tril[1]],
+                                   rtol=1e-4)
+
+        # hermitian-filled output: compare everything
+        full = np.concatenate(full_out, axis=0)
+        np.testing.assert_allclose(full, expected, rtol=1e-4)
+
+        # matrix -> storage -> matrix round trip == direct fill
+        rt = np.concatenate(storage_rt, axis=0)
+        np.testing.assert_allclose(rt, full, rtol=1e-4, atol=1e-2)
