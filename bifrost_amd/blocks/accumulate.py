@@ -5,7 +5,6 @@ from copy import deepcopy
 
 import numpy as np
 
-from bifrost_amd.ndarray import copy_array
 from bifrost_amd.pipeline import TransformBlock
 
 __all__ = ["AccumulateBlock", "accumulate"]

@@ -15,7 +15,6 @@ def unpack(src, dst, align_msb=False):
 
 
 def unpack_new(src, dst_dtype, align_msb=False):
-    from bifrost_amd.ndarray import empty_like
     src_bf = asarray(src)
     from bifrost_amd.ndarray import ndarray
     dst = ndarray(shape=src_bf.shape, dtype=dst_dtype, space=src_bf.bf.space)
