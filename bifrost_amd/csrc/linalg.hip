@@ -456,10 +456,10 @@ void cherk_ci8_mfma_kernel(long n, long k, long nbatch, float alpha,
 //   4 = schedule 3 + double-buffered stg: loads for slab s+2 issue BEFORE
 //       the ds_writes of slab s+1, so the writes wait vmcnt(4) (the old
 //       loads only) and the new loads get a slab of extra latency slack
-template <int NHALF, int SCHED = 2>
+template <int NHALF, int SCHED = 5>
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(
-    NHALF != 1 ? 2 : (SCHED >= 3 ? 3 : CHERK_RS_W1))))
+    NHALF != 1 ? 2 : ((SCHED == 3 || SCHED == 4) ? 3 : CHERK_RS_W1))))
 void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
                               const signed char* __restrict__ a, long lda,
                               long a_b, float beta, f2* __restrict__ c,
@@ -588,6 +588,25 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
                             acc[ta][tb] =
                                 __builtin_amdgcn_mfma_i32_16x16x64_i8(
                                     fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+                } else if (SCHED == 5) {
+                    // two-row-ahead fa pipeline: each group's fragment
+                    // was issued TWO groups earlier (lgkmcnt distance 4)
+                    v4i fb[4];
+                    for (int tb = 0; tb < 4; ++tb)
+                        fb[tb] = frag(bJ, 4 * wc + tb);
+                    v4i fa0 = frag(bI, 4 * wr);
+                    v4i fa1 = frag(bI, 4 * wr + 1);
+#pragma unroll
+                    for (int ta = 0; ta < 4; ++ta) {
+                        v4i fa2;
+                        if (ta < 2) fa2 = frag(bI, 4 * wr + ta + 2);
+                        for (int tb = 0; tb < 4; ++tb)
+                            acc[ta][tb] =
+                                __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                                    fa0, fb[tb], acc[ta][tb], 0, 0, 0);
+                        fa0 = fa1;
+                        fa1 = fa2;
+                    }
                 } else {  // one-row-pipelined burst
                     v4i fb[4];
                     for (int tb = 0; tb < 4; ++tb)
@@ -1409,7 +1428,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 // over barrier amortization; measured 1.19 vs 1.11 Gsamp/s)
                 int nhalf = bkenv ? atoi(bkenv) / 64 : 1;
                 const char* schenv = getenv("BIFROST_CHERK_SCHED");
-                int sched = schenv ? atoi(schenv) : 2;
+                int sched = schenv ? atoi(schenv) : 5;
                 auto launch_rs1 = [&](auto kern) {
                     hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, n,
                                        k, nbatch, (float)alpha,
@@ -1427,6 +1446,8 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                     launch_rs1(cherk_ci8_mfma_rs_kernel<1, 3>);
                 else if (sched == 4)
                     launch_rs1(cherk_ci8_mfma_rs_kernel<1, 4>);
+                else if (sched == 5)
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 5>);
                 else
                     launch_rs1(cherk_ci8_mfma_rs_kernel<1, 2>);
                 BF_CHECK_HIP(hipGetLastError());
