@@ -8,6 +8,8 @@ from bifrost_amd.blocks.binary_io import (BinaryFileReadBlock,  # noqa: F401
 from bifrost_amd.blocks.convert_visibilities import (  # noqa: F401
     ConvertVisibilitiesBlock, convert_visibilities)
 from bifrost_amd.blocks.copy import CopyBlock, copy  # noqa: F401
+from bifrost_amd.blocks.dada_file import (DadaFileReadBlock,  # noqa: F401
+                                          read_dada_file)
 from bifrost_amd.blocks.guppi_raw import GuppiRawSourceBlock, read_guppi_raw  # noqa: F401
 from bifrost_amd.blocks.serialize import (DeserializeBlock,  # noqa: F401
                                           SerializeBlock, deserialize,

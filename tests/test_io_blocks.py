@@ -261,3 +261,48 @@ class TestSigprocBlocks:
         assert shdr["_tensor"]["labels"] == ["time", "pol", "freq"]
         assert shdr["source_name"] == "J0000+0000"
         assert shdr["telescope"] == "GBT"
+
+
+# ---------------------------------------------------------------------------
+# DADA files
+
+def _write_dada(fname, hdr_pairs, data):
+    with open(fname, "wb") as f:
+        txt = "".join("%s %s\n" % kv for kv in hdr_pairs.items())
+        f.write(txt.encode().ljust(4096, b"\0"))
+        data.tofile(f)
+
+
+class TestDadaFile:
+    def test_read_pipeline(self, tmp_path):
+        nframe, nchan = 12, 16
+        rng = np.random.RandomState(7)
+        data = rng.standard_normal((nframe, nchan)).astype(np.float32)
+        f1 = str(tmp_path / "cap_0000000000.000000.dada")
+        f2 = str(tmp_path / "cap_0000000001.000000.dada")
+        hdr = {"NCHAN": nchan, "NBIT": 32, "TSAMP": "1.0"}
+        _write_dada(f1, hdr, data[:6])
+        _write_dada(f2, hdr, data[6:])
+
+        def header_callback(dada_hdr):
+            nch = int(dada_hdr["NCHAN"])
+            return {
+                "name": "dada-test",
+                "time_tag": 0,
+                "_tensor": {
+                    "dtype": "f32",
+                    "shape": [-1, nch],
+                    "labels": ["time", "freq"],
+                    "scales": [[0, 1], [0, 1]],
+                    "units": [None, None],
+                },
+            }
+
+        out = []
+        with bf.Pipeline() as pipe:
+            src = bf.blocks.read_dada_file([f1], header_callback,
+                                           gulp_nframe=1)
+            CollectBlock(src, out)
+            pipe.run()
+        got = np.concatenate(out, axis=0)
+        np.testing.assert_array_equal(got, data)  # spans both files
