@@ -656,6 +656,16 @@ void cherk_ci8_mfma_rs_kernel(long n, long k, long nbatch, float alpha,
                         if (s + 2 < nslab) load_into(fill);
                         write_from(wrbuf, consume);
                     }
+                } else if (SCHED == 6) {
+                    // DIAGNOSTIC (wrong results): compute-only ceiling —
+                    // no staging, every slab re-reads buffer 0
+                    if (!skip_all) burst(bI, bJ);
+                } else if (SCHED == 7) {
+                    // DIAGNOSTIC (wrong results): staging-only floor
+                    if (s + 1 < nslab) {
+                        write_from(wrbuf, consume);
+                        if (s + 2 < nslab) load_into(fill);
+                    }
                 } else {
                     // SCHED 2/3: burst -> writes -> loads (one stg set)
                     if (!skip_all) burst(bI, bJ);
@@ -1448,6 +1458,10 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                     launch_rs1(cherk_ci8_mfma_rs_kernel<1, 4>);
                 else if (sched == 5)
                     launch_rs1(cherk_ci8_mfma_rs_kernel<1, 5>);
+                else if (sched == 6)  // diagnostic: wrong results
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 6>);
+                else if (sched == 7)  // diagnostic: wrong results
+                    launch_rs1(cherk_ci8_mfma_rs_kernel<1, 7>);
                 else
                     launch_rs1(cherk_ci8_mfma_rs_kernel<1, 2>);
                 BF_CHECK_HIP(hipGetLastError());
