@@ -51,5 +51,10 @@ class BlockChainer(object):
 
     @property
     def views(self):
-        raise NotImplementedError("ring views are not yet implemented on "
-                                  "this backend (see DESIGN.md)")
+        import bifrost_amd.views as _views_module
+        return _ChainProxy(self, _views_module)
+
+    def print_header(self, *args, **kwargs):
+        from bifrost_amd.blocks.print_header import print_header
+        block = print_header(self.last_block, *args, **kwargs)
+        return block

@@ -508,40 +508,60 @@ BFstatus bfRingSpanAcquire(BFrspan* span, BFrsequence sequence,
     BF_ASSERT(ring->buf, BF_STATUS_INVALID_STATE);
     BF_ASSERT(size <= ring->ghost, BF_STATUS_INVALID_ARGUMENT);
     Sequence* seq = sequence->seq.get();
-    BFoffset abs = seq->begin + offset;
-    BFsize overwritten = 0;
+    // Reference semantics (src/ring_impl.cpp:633-701): return whatever
+    // part of the REQUESTED window [req_begin, req_begin+size) survives —
+    // begin = max(req_begin, tail), possibly a zero-length span when the
+    // window was fully overwritten.  NEVER jump beyond the request: the
+    // caller's frame-offset bookkeeping (ring2.read 'offset += stride')
+    // depends on each acquire covering exactly its window, and the
+    // pipeline's _on_skip/zero-span paths handle the rest.  (The earlier
+    // fast-forward-to-tail behaviour re-delivered fresh data for stale
+    // offsets: an unguaranteed reader chain then REPLAYS the stream ~2x
+    // per hop — exponential in pipeline depth.)
+    BFoffset req_begin = seq->begin + offset;
+    BFoffset req_end = req_begin + size;
+    if (sequence->guarantee && req_begin > sequence->guard) {
+        // move the guarantee forward so writers can progress
+        sequence->guard = req_begin;
+        ring->cv.notify_all();
+    }
     for (;;) {
-        if (seq->ended && abs >= seq->end) return BF_STATUS_END_OF_DATA;
-        // A globally-ended ring can never deliver more: clamp to head.
-        if (ring->writing_ended && !seq->ended) {
-            if (abs >= ring->head) return BF_STATUS_END_OF_DATA;
-        }
-        BFsize want = size;
-        if (seq->ended && abs + want > seq->end) want = seq->end - abs;
-        if (ring->writing_ended && !seq->ended && abs + want > ring->head)
-            want = ring->head - abs;
-        if (ring->head >= abs + want && want > 0) {
-            // Data (or final partial gulp) available.
-            if (!sequence->guarantee && abs < ring->tail()) {
-                // Overwritten: skip forward in whole-gulp steps to keep the
-                // caller's frame alignment (pipeline skip logic relies on
-                // frame-aligned offsets, python ring2.py ReadSpan).
-                BFoffset deficit = ring->tail() - abs;
-                BFoffset steps = (deficit + size - 1) / size;
-                abs += steps * size;
-                overwritten = 0;
+        if (seq->ended && req_begin >= seq->end) return BF_STATUS_END_OF_DATA;
+        if (ring->writing_ended && !seq->ended && req_begin >= ring->head)
+            return BF_STATUS_END_OF_DATA;
+        BFoffset avail_begin = std::max<BFoffset>(req_begin, ring->tail());
+        bool finished = seq->ended ||
+                        (ring->writing_ended && !seq->ended);
+        if ((BFdelta)(ring->head - avail_begin) >=
+                (BFdelta)(req_end - avail_begin) ||
+            finished) {
+            BFoffset begin = std::max<BFoffset>(req_begin, ring->tail());
+            BFdelta ssize = (BFdelta)(req_end - begin);
+            if (ssize < 0) ssize = 0;
+            if (seq->ended) {
+                if (begin >= seq->end) return BF_STATUS_END_OF_DATA;
+                ssize = std::min<BFdelta>(ssize, (BFdelta)(seq->end - begin));
+            } else if (ring->writing_ended) {
+                if (begin >= ring->head) return BF_STATUS_END_OF_DATA;
+                ssize = std::min<BFdelta>(ssize,
+                                          (BFdelta)(ring->head - begin));
+            } else if ((BFdelta)(ring->head - begin) < ssize) {
+                // window partially overwritten but head not there yet
+                ring->cv.wait(lk);
                 continue;
             }
-            BFstatus st = ring->refresh_ghost(abs, want);
-            if (st != BF_STATUS_SUCCESS) return st;
-            sequence->guard = abs;
+            if (ssize > 0) {
+                BFstatus st = ring->refresh_ghost(begin, (BFsize)ssize);
+                if (st != BF_STATUS_SUCCESS) return st;
+            }
+            sequence->guard = begin;
             auto* rs = new BFrspan_impl();
             rs->is_write = false;
             rs->ring = ring;
             rs->rseq = sequence;
-            rs->begin = abs;
-            rs->size = want;
-            rs->size_overwritten = overwritten;
+            rs->begin = begin;
+            rs->size = (BFsize)ssize;
+            rs->size_overwritten = 0;
             *span = rs;
             return BF_STATUS_SUCCESS;
         }
