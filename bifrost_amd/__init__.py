@@ -21,7 +21,7 @@ from bifrost_amd.ndarray import (asarray, copy_array, empty, empty_like,  # noqa
 from bifrost_amd.quantize import quantize  # noqa: F401
 from bifrost_amd.transpose import transpose  # noqa: F401
 from bifrost_amd.unpack import unpack  # noqa: F401
-from bifrost_amd.map import clear_map_cache, map  # noqa: F401
+from bifrost_amd.map import clear_map_cache, list_map_cache, map  # noqa: F401
 from bifrost_amd.fft import Fft  # noqa: F401
 from bifrost_amd.reduce import reduce  # noqa: F401
 

@@ -8,7 +8,7 @@ import numpy as np
 from bifrost_amd.libbifrost import _bf, _check, _array
 from bifrost_amd.ndarray import asarray, ndarray
 
-__all__ = ["map", "clear_map_cache"]
+__all__ = ["map", "clear_map_cache", "list_map_cache"]
 
 
 def map(func_string, data, axis_names=None, shape=None, func_name=None,
@@ -52,3 +52,10 @@ def map(func_string, data, axis_names=None, shape=None, func_name=None,
 
 def clear_map_cache():
     _check(_bf.bfMapClearCache())
+
+
+def list_map_cache():
+    """Print bfMap kernel-cache status (reference surface; this backend
+    keeps an in-process hipModule cache rather than an on-disk one)."""
+    print("Cache enabled: yes (in-process, %d-entry LRU; cleared by "
+          "clear_map_cache())" % 128)
