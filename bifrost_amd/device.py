@@ -39,3 +39,40 @@ def get_stream():
     s = ctypes.c_void_p()
     _check(_bf.bfStreamGet(ctypes.byref(s)))
     return s.value if s.value is not None else 0
+
+
+class ExternalStream(object):
+    """Context manager to run bifrost ops on a stream created outside
+    (e.g. a torch.cuda.Stream): the original stream is restored on exit.
+
+        with bf.device.ExternalStream(torch.cuda.current_stream()):
+            ...
+    """
+
+    def __init__(self, stream):
+        self._stream = stream
+
+    def use(self):
+        self._orig_stream = get_stream()
+        stream = getattr(self._stream, "cuda_stream", None)  # torch
+        if stream is None:
+            stream = getattr(self._stream, "ptr", None)  # cupy
+        if stream is None:
+            stream = getattr(self._stream, "handle", None)  # pycuda
+        if stream is None:
+            stream = self._stream
+        set_stream(stream)
+
+    def __enter__(self):
+        self.use()
+        return self
+
+    def __exit__(self, t, v, tb):
+        set_stream(self._orig_stream)
+        del self._orig_stream
+
+    def __del__(self):
+        try:
+            set_stream(self._orig_stream)
+        except AttributeError:
+            pass
