@@ -134,3 +134,36 @@ def test_space_type_conversion():
         d = c.astype(dtype).copy(space="system")
         np.testing.assert_equal(b, np.asarray(d), err_msg="cf32->%s" %
                                 dtype)
+
+
+def test_external_stream_torch():
+    """Run a bifrost op on a torch stream via device.ExternalStream."""
+    import torch
+    from bifrost_amd import device
+
+    s = torch.cuda.Stream()
+    a = bf.asarray(np.arange(128, dtype=np.float32), space="cuda")
+    c = bf.zeros((128,), dtype="f32", space="cuda")
+    orig = device.get_stream()
+    with device.ExternalStream(s):
+        assert device.get_stream() == s.cuda_stream
+        bf.map("c = a * 3", {"c": c, "a": a})
+        device.stream_synchronize()
+    assert device.get_stream() == orig
+    np.testing.assert_array_equal(np.asarray(c.copy("system")),
+                                  np.arange(128) * 3)
+
+
+def test_torch_buffer_interop():
+    """bf.ndarray over a torch allocation (the bench.py bridge)."""
+    import torch
+    t = torch.zeros((16, 2), dtype=torch.float32, device="cuda")
+    v = bf.ndarray(space="cuda", shape=(16,), dtype="cf32",
+                   buffer=t.data_ptr())
+    bf.map("v = Complex<float>(7, -3)", {"v": v})
+    from bifrost_amd import device
+    device.stream_synchronize()
+    torch.cuda.synchronize()
+    got = t.cpu().numpy()
+    np.testing.assert_array_equal(got[:, 0], np.full(16, 7.0))
+    np.testing.assert_array_equal(got[:, 1], np.full(16, -3.0))
