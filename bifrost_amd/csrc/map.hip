@@ -22,6 +22,8 @@
 #include <hip/hiprtc.h>
 
 #include <cctype>
+#include <cstdio>
+#include <sys/stat.h>
 #include <cstring>
 #include <list>
 #include <mutex>
@@ -211,6 +213,84 @@ void cache_evict_locked() {
     }
 }
 
+// ---- on-disk code cache ----------------------------------------------------
+// Like the reference's ~/.bifrost map kernel disk cache: compiled code
+// objects keyed by an FNV-1a hash of the generated source, so cold
+// processes skip hipRTC entirely.  Disable with BIFROST_NO_DISK_CACHE=1.
+
+std::string disk_cache_dir() {
+    static std::string dir = [] {
+        if (getenv("BIFROST_NO_DISK_CACHE")) return std::string();
+        const char* home = getenv("HOME");
+        if (!home || !*home) return std::string();
+        std::string d = std::string(home) + "/.bifrost_amd";
+        mkdir(d.c_str(), 0755);
+        d += "/mapcache_gfx950";
+        mkdir(d.c_str(), 0755);
+        return d;
+    }();
+    return dir;
+}
+
+std::string src_hash_hex(const std::string& src) {
+    unsigned long long h = 1469598103934665603ULL;  // FNV-1a 64
+    for (unsigned char c : src) {
+        h ^= c;
+        h *= 1099511628211ULL;
+    }
+    char buf[20];
+    std::snprintf(buf, sizeof(buf), "%016llx", h);
+    return buf;
+}
+
+bool disk_cache_load(const std::string& src, std::vector<char>* code) {
+    std::string dir = disk_cache_dir();
+    if (dir.empty()) return false;
+    std::string path = dir + "/" + src_hash_hex(src) + ".hsaco";
+    FILE* f = std::fopen(path.c_str(), "rb");
+    if (!f) return false;
+    std::fseek(f, 0, SEEK_END);
+    long n = std::ftell(f);
+    std::fseek(f, 0, SEEK_SET);
+    if (n <= 0) { std::fclose(f); return false; }
+    code->resize((size_t)n);
+    bool ok = std::fread(code->data(), 1, (size_t)n, f) == (size_t)n;
+    std::fclose(f);
+    return ok;
+}
+
+void disk_cache_store(const std::string& src,
+                      const std::vector<char>& code) {
+    std::string dir = disk_cache_dir();
+    if (dir.empty()) return;
+    std::string path = dir + "/" + src_hash_hex(src) + ".hsaco";
+    std::string tmp = path + ".tmp";
+    FILE* f = std::fopen(tmp.c_str(), "wb");
+    if (!f) return;
+    bool ok = std::fwrite(code.data(), 1, code.size(), f) == code.size();
+    std::fclose(f);
+    if (ok) std::rename(tmp.c_str(), path.c_str());
+    else std::remove(tmp.c_str());
+}
+
+BFstatus load_module(const std::vector<char>& code, const std::string& src,
+                     CacheEntry* out) {
+    CacheEntry e;
+    if (hipModuleLoadData(&e.module, code.data()) != hipSuccess)
+        return BF_STATUS_DEVICE_ERROR;
+    if (hipModuleGetFunction(&e.func, e.module, "bfmap_kernel") !=
+        hipSuccess) {
+        hipModuleUnload(e.module);
+        return BF_STATUS_DEVICE_ERROR;
+    }
+    std::lock_guard<std::mutex> lk(g_cache_mutex);
+    g_cache[src] = e;
+    g_cache_order.push_back(src);
+    cache_evict_locked();
+    *out = e;
+    return BF_STATUS_SUCCESS;
+}
+
 BFstatus compile_and_cache(const std::string& src, CacheEntry* out) {
     {
         std::lock_guard<std::mutex> lk(g_cache_mutex);
@@ -219,6 +299,12 @@ BFstatus compile_and_cache(const std::string& src, CacheEntry* out) {
             *out = it->second;
             return BF_STATUS_SUCCESS;
         }
+    }
+    {
+        std::vector<char> code;
+        if (disk_cache_load(src, &code) &&
+            load_module(code, src, out) == BF_STATUS_SUCCESS)
+            return BF_STATUS_SUCCESS;
     }
     hiprtcProgram prog;
     if (hiprtcCreateProgram(&prog, src.c_str(), "bfmap.hip", 0, nullptr,
@@ -244,19 +330,8 @@ BFstatus compile_and_cache(const std::string& src, CacheEntry* out) {
     hiprtcGetCode(prog, code.data());
     hiprtcDestroyProgram(&prog);
 
-    CacheEntry e;
-    if (hipModuleLoadData(&e.module, code.data()) != hipSuccess)
-        return BF_STATUS_DEVICE_ERROR;
-    if (hipModuleGetFunction(&e.func, e.module, "bfmap_kernel") != hipSuccess) {
-        hipModuleUnload(e.module);
-        return BF_STATUS_DEVICE_ERROR;
-    }
-    std::lock_guard<std::mutex> lk(g_cache_mutex);
-    g_cache[src] = e;
-    g_cache_order.push_back(src);
-    cache_evict_locked();
-    *out = e;
-    return BF_STATUS_SUCCESS;
+    disk_cache_store(src, code);
+    return load_module(code, src, out);
 }
 
 }  // namespace
