@@ -257,9 +257,28 @@ class ndarray(np.ndarray):
                 a = asarray(a)
             a.bf.dtype = dtype_bf
             return a
-        # device arrays: JIT an elementwise conversion through bfMap.
-        # The package attribute 'map' is rebound to the function by
-        # __init__, so fetch the real module from sys.modules.
+        # device arrays: packed ci4 goes through the native unpack/
+        # quantize kernels (bfMap cannot address sub-byte elements) ...
+        if self.bf.dtype == DataType("ci4") or dtype_bf == DataType("ci4"):
+            # (package attrs 'unpack'/'quantize' are rebound to functions
+            # by __init__, so fetch the real modules)
+            import importlib
+            _quantize = importlib.import_module("bifrost_amd.quantize")
+            _unpack = importlib.import_module("bifrost_amd.unpack")
+            a = ndarray(shape=self.shape, dtype=dtype_bf,
+                        space=self.bf.space)
+            if self.bf.dtype == DataType("ci4"):
+                if dtype_bf in (DataType("ci8"), DataType("cf32"),
+                                DataType("cf64")):
+                    _unpack.unpack(self, a)
+                    return a
+                tmp = self.astype("cf32")
+                return tmp.astype(dtype)
+            _quantize.quantize(self, a, scale=1.0)
+            return a
+        # ... everything else JITs an elementwise conversion through
+        # bfMap.  The package attribute 'map' is rebound to the function
+        # by __init__, so fetch the real module from sys.modules.
         import importlib
         bf_map = importlib.import_module("bifrost_amd.map")
         a = ndarray(shape=self.shape, dtype=dtype_bf, space=self.bf.space)

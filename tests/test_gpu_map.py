@@ -209,3 +209,37 @@ class TestIndexedForm:
                axis_names=("i",))
         np.testing.assert_array_equal(np.asarray(b.copy("system")),
                                       np.arange(10))
+
+
+class TestCi4Astype:
+    """Device astype for packed ci4 routes through unpack/quantize."""
+
+    def test_ci4_to_ci8(self):
+        raw = np.array([(0x10,), (0x32,), (0xBA,)],
+                       dtype=bf.DataType.ci4)
+        a = bf.asarray(bf.ndarray(raw), space="cuda")
+        out = a.astype("ci8").copy("system")
+        got = np.asarray(out)
+        np.testing.assert_array_equal(got["re"], [0, 2, -6])
+        np.testing.assert_array_equal(got["im"], [1, 3, -5])
+
+    def test_ci4_to_cf32(self):
+        raw = np.array([(0x10,), (0x32,)], dtype=bf.DataType.ci4)
+        a = bf.asarray(bf.ndarray(raw), space="cuda")
+        got = np.asarray(a.astype("cf32").copy("system"))
+        np.testing.assert_array_equal(got, [0 + 1j, 2 + 3j])
+
+    def test_cf32_to_ci4(self):
+        c = np.array([1 + 2j, -3 - 4j], dtype=np.complex64)
+        a = bf.asarray(c, space="cuda")
+        out = a.astype("ci4").copy("system")
+        # ci4 packs re in the HIGH nibble (quantize/linalg convention):
+        # 1+2j -> 0x12; -3-4j -> 0xDC
+        np.testing.assert_array_equal(np.asarray(out)["re_im"],
+                                      [0x12, 0xDC])
+        # NOTE: unpack decodes re from the LOW nibble (the reference's own
+        # convention asymmetry between quantize and unpack, DESIGN.md §4),
+        # so the quantize->unpack round trip swaps re/im — faithfully.
+        back = np.asarray(a.astype("ci4").astype("cf32").copy("system"))
+        np.testing.assert_array_equal(back, np.array([2 + 1j, -4 - 3j],
+                                                     dtype=np.complex64))
