@@ -1279,6 +1279,152 @@ __global__ __launch_bounds__(256) void beamform_fast_kernel(
     }
 }
 
+/* ------------- bf16-split MFMA beamformer (cf32 W x ci8 X) -------------- */
+// Same contract as beamform_fast_kernel, on the bf16 matrix cores: the
+// cf32 weights split into hi+lo bf16 (W = W_hi + W_lo captures 16
+// mantissa bits; the int8 voltages are EXACT in bf16), so each of the 4
+// real GEMM planes runs as two v_mfma_f32_16x16x32_bf16 accumulations —
+// an ~1.25 PF effective ceiling vs the ~0.16 PF packed-f32 VALU path.
+// Fragment layout matches the i8 MFMA (lane&15 = A-row/B-col, 8
+// k-elements per lane at (lane>>4)*8; any consistent k-permutation of A
+// and B cancels); D: col = lane&15 (time), row = (lane>>4)*4 + reg
+// (beam).  Requires k%64==0, nn%64==0, beam chunk = 16*NBT.
+typedef __bf16 bf16_t;
+typedef bf16_t v8bf __attribute__((ext_vector_type(8)));
+typedef float v4f __attribute__((ext_vector_type(4)));
+
+template <int NBT, int JT = 2>  // JT time-tiles of 16 per wave
+__global__ __launch_bounds__(256) void beamform_mfma_kernel(
+    long nn, long k, long nbatch, float alpha, const f2* __restrict__ w,
+    long ldw, long w_b, const signed char* __restrict__ x, long ldx,
+    long x_b, float beta, f2* __restrict__ c, long c_row, long c_b,
+    long i0) {
+    // padded rows (72 = 64 + 8) keep v8bf reads 16B-aligned and spread
+    // the 16-row fragment reads over distinct banks
+    __shared__ bf16_t swp[NBT][4][16][72];   // [beam-tile][hr,hi,lr,li]
+    __shared__ bf16_t sxp[4 * JT][2][16][72];  // [time-tile][re,im]
+    int tid = threadIdx.x;
+    int wave = tid >> 6, lane = tid & 63;
+    int row16 = lane & 15, kblk = lane >> 4;
+    const int TW = 64 * JT;  // times per workgroup
+    for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
+        const f2* wb = w + batch * w_b;
+        const signed char* xb = x + 2 * (batch * x_b);
+        f2* cb = c + batch * c_b;
+        for (long j0 = (long)blockIdx.x * TW; j0 < nn;
+             j0 += (long)gridDim.x * TW) {
+            v4f accr[NBT][JT], acci[NBT][JT];
+            for (int t = 0; t < NBT; ++t)
+                for (int u = 0; u < JT; ++u) {
+                    accr[t][u] = v4f{0.f, 0.f, 0.f, 0.f};
+                    acci[t][u] = v4f{0.f, 0.f, 0.f, 0.f};
+                }
+            for (long k0 = 0; k0 < k; k0 += 64) {
+                for (int idx = tid; idx < NBT * 16 * 16; idx += 256) {
+                    int r = idx / 16, q = idx % 16;  // row, 4-k group
+                    const f2* src = wb + (i0 + r) * ldw + k0 + 4 * q;
+                    int bt = r >> 4, rr = r & 15;
+                    for (int e = 0; e < 4; ++e) {
+                        f2 v = src[e];
+                        bf16_t hr = (bf16_t)v.x;
+                        bf16_t hi = (bf16_t)v.y;
+                        swp[bt][0][rr][4 * q + e] = hr;
+                        swp[bt][1][rr][4 * q + e] = hi;
+                        swp[bt][2][rr][4 * q + e] =
+                            (bf16_t)(v.x - (float)hr);
+                        swp[bt][3][rr][4 * q + e] =
+                            (bf16_t)(v.y - (float)hi);
+                    }
+                }
+                for (int idx = tid; idx < TW * 8; idx += 256) {
+                    int r = idx >> 3, q = idx & 7;  // time row, 8-k group
+                    const signed char* src =
+                        xb + 2 * ((j0 + r) * ldx + k0 + 8 * q);
+                    int jt = r >> 4, rr = r & 15;
+                    signed char buf[16];
+                    __builtin_memcpy(buf, src, 16);
+                    // int8 values are exact in bf16, so the conversion is
+                    // a float-bits truncation (low mantissa bits are 0);
+                    // pack pairs into 32-bit LDS writes
+                    unsigned* pr = (unsigned*)&sxp[jt][0][rr][8 * q];
+                    unsigned* pi = (unsigned*)&sxp[jt][1][rr][8 * q];
+                    for (int e = 0; e < 4; ++e) {
+                        unsigned r0 = __builtin_bit_cast(
+                            unsigned, (float)buf[4 * e + 0]) >> 16;
+                        unsigned i0b = __builtin_bit_cast(
+                            unsigned, (float)buf[4 * e + 1]) >> 16;
+                        unsigned r1 = __builtin_bit_cast(
+                            unsigned, (float)buf[4 * e + 2]) >> 16;
+                        unsigned i1 = __builtin_bit_cast(
+                            unsigned, (float)buf[4 * e + 3]) >> 16;
+                        pr[e] = r0 | (r1 << 16);
+                        pi[e] = i0b | (i1 << 16);
+                    }
+                }
+                __syncthreads();
+                for (int kc = 0; kc < 2; ++kc) {
+                    int kof = 32 * kc + 8 * kblk;
+                    v8bf xr[JT], xi[JT], xni[JT];
+                    for (int u = 0; u < JT; ++u) {
+                        int jt = wave * JT + u;
+                        xr[u] = *(const v8bf*)&sxp[jt][0][row16][kof];
+                        xi[u] = *(const v8bf*)&sxp[jt][1][row16][kof];
+                        xni[u] = -xi[u];  // sign flip, packed ops
+                    }
+                    for (int bt = 0; bt < NBT; ++bt) {
+                        v8bf whr = *(const v8bf*)&swp[bt][0][row16][kof];
+                        v8bf whi = *(const v8bf*)&swp[bt][1][row16][kof];
+                        v8bf wlr = *(const v8bf*)&swp[bt][2][row16][kof];
+                        v8bf wli = *(const v8bf*)&swp[bt][3][row16][kof];
+                        for (int u = 0; u < JT; ++u) {
+                            // Yr += (Whr+Wlr)·Xr − (Whi+Wli)·Xi
+                            accr[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    whr, xr[u], accr[bt][u], 0, 0, 0);
+                            accr[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    wlr, xr[u], accr[bt][u], 0, 0, 0);
+                            accr[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    whi, xni[u], accr[bt][u], 0, 0, 0);
+                            accr[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    wli, xni[u], accr[bt][u], 0, 0, 0);
+                            // Yi += (Whr+Wlr)·Xi + (Whi+Wli)·Xr
+                            acci[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    whr, xi[u], acci[bt][u], 0, 0, 0);
+                            acci[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    wli, xr[u], acci[bt][u], 0, 0, 0);
+                            acci[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    whi, xr[u], acci[bt][u], 0, 0, 0);
+                            acci[bt][u] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    wlr, xi[u], acci[bt][u], 0, 0, 0);
+                        }
+                    }
+                }
+                __syncthreads();
+            }
+            for (int u = 0; u < JT; ++u) {
+                long j = j0 + (wave * JT + u) * 16 + row16;
+                for (int bt = 0; bt < NBT; ++bt) {
+                    for (int r = 0; r < 4; ++r) {
+                        long beam = i0 + bt * 16 + kblk * 4 + r;
+                        f2 prev =
+                            beta != 0.f ? cb[beam * c_row + j] : f2{};
+                        cb[beam * c_row + j] =
+                            f2{alpha * accr[bt][u][r] + beta * prev.x,
+                               alpha * acci[bt][u][r] + beta * prev.y};
+                    }
+                }
+            }
+        }
+    }
+}
+
 // Each thread owns TWO consecutive time samples so every W read from LDS
 // feeds two complex MACs (the kernel is otherwise LDS-read-bound).
 template <typename LoadW, typename LoadX, int MTILE>
@@ -1562,6 +1708,33 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
         !conj_b && m <= 1024) {
         dim3 grid(cap_grid((nn + 511) / 512, 4096), cap_grid(nbatch, 65535));
         bool fast_ok = (k % 64 == 0) && (nn % 2 == 0) && nn >= 2;
+        // bf16-split MFMA path (cf32 W, ci8 X, full tiles): ~1.25 PF
+        // effective ceiling vs the f32 VALU kernels.  Opt out with
+        // BIFROST_BEAM=valu.
+        const char* beam_sel = getenv("BIFROST_BEAM");
+        bool want_mfma = !(beam_sel && strcmp(beam_sel, "valu") == 0);
+        if (want_mfma && a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI8 &&
+            k % 64 == 0 && nn % 128 == 0 && m % 16 == 0 && nn > 0) {
+            dim3 mgrid(cap_grid(nn / 128, 4096), cap_grid(nbatch, 65535));
+            for (long i0 = 0; i0 < m;) {
+                long chunk = std::min<long>(64, m - i0);
+#define BEAM_MFMA_CASE(NBT)                                                   \
+    hipLaunchKernelGGL((beamform_mfma_kernel<NBT>), mgrid, dim3(256), 0,      \
+                       stream, nn, k, nbatch, (float)alpha, (const f2*)a,     \
+                       a_i, a_b, (const signed char*)b, b_j, b_b,             \
+                       (float)beta, (f2*)c, c_row, c_b, i0)
+                switch (chunk / 16) {
+                case 4: BEAM_MFMA_CASE(4); break;
+                case 3: BEAM_MFMA_CASE(3); break;
+                case 2: BEAM_MFMA_CASE(2); break;
+                default: BEAM_MFMA_CASE(1); chunk = 16; break;
+                }
+#undef BEAM_MFMA_CASE
+                BF_CHECK_HIP(hipGetLastError());
+                i0 += chunk;
+            }
+            return BF_STATUS_SUCCESS;
+        }
         for (long i0 = 0; i0 < m; i0 += 16) {
             if (fast_ok && i0 + 16 <= m) {
 #define BEAMF_CASE(LW, LX)                                                    \

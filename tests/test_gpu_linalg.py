@@ -225,3 +225,60 @@ class TestMatMulAB:
         linalg.matmul(1, H(bg), H(ag), 0, c)
         np.testing.assert_allclose(np.asarray(c.copy("system")), c_gold,
                                    RTOL, ATOL)
+
+
+class TestBeamformerMFMA:
+    """The bf16-split MFMA path (cf32 W x ci8 X, k%64==0, ntime%128==0):
+    parity at each beam-chunk shape, plus agreement with the VALU kernel."""
+
+    @pytest.mark.parametrize("nbeam", [16, 48, 64])
+    def test_mfma_parity(self, linalg, nbeam):
+        run_beam(linalg, 256, nbeam, 32, 8)
+
+    def test_beta_accumulate(self, linalg):
+        np.random.seed(77)
+        ntime, nbeam, nstand, nchan = 128, 32, 32, 4
+        x8 = ((np.random.random((ntime, nchan, nstand * 2, 2)) * 2 - 1)
+              * 127).astype(np.int8)
+        x = x8.astype(np.float32).view(np.complex64) \
+            .reshape(ntime, nchan, nstand * 2)
+        w = np.random.standard_normal(
+            (nbeam, nchan, nstand * 2, 2)).astype(np.float32) \
+            .view(np.complex64).reshape(nbeam, nchan, nstand * 2)
+        gold = np.matmul(w.transpose(1, 0, 2), x.transpose(1, 2, 0))
+        xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                   .reshape(ntime, nchan, nstand * 2)),
+                        space="cuda")
+        wb = bf.asarray(w, space="cuda")
+        c = bf.zeros_like(gold, space="cuda")
+        linalg.matmul(0.5, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0),
+                      0, c)
+        linalg.matmul(0.5, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0),
+                      1, c)
+        np.testing.assert_allclose(np.asarray(c.copy("system")), gold,
+                                   1e-4, 1e-2)
+
+    def test_matches_valu_kernel(self, linalg, monkeypatch):
+        import os
+        np.random.seed(78)
+        ntime, nbeam, nstand, nchan = 128, 64, 32, 4
+        x8 = ((np.random.random((ntime, nchan, nstand * 2, 2)) * 2 - 1)
+              * 127).astype(np.int8)
+        w = np.random.standard_normal(
+            (nbeam, nchan, nstand * 2, 2)).astype(np.float32) \
+            .view(np.complex64).reshape(nbeam, nchan, nstand * 2)
+        xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                   .reshape(ntime, nchan, nstand * 2)),
+                        space="cuda")
+        wb = bf.asarray(w, space="cuda")
+        c1 = bf.zeros((nchan, nbeam, ntime), dtype="cf32", space="cuda")
+        c2 = bf.zeros((nchan, nbeam, ntime), dtype="cf32", space="cuda")
+        linalg.matmul(1, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0),
+                      0, c1)
+        monkeypatch.setenv("BIFROST_BEAM", "valu")
+        linalg.matmul(1, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0),
+                      0, c2)
+        a = np.asarray(c1.copy("system"))
+        b = np.asarray(c2.copy("system"))
+        np.testing.assert_allclose(a, b, rtol=2e-4,
+                                   atol=1e-3 * np.abs(b).max())
