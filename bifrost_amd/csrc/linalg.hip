@@ -1293,7 +1293,7 @@ typedef __bf16 bf16_t;
 typedef bf16_t v8bf __attribute__((ext_vector_type(8)));
 typedef float v4f __attribute__((ext_vector_type(4)));
 
-template <int NBT, int JT = 2>  // JT time-tiles of 16 per wave
+template <int NBT, int XT = 0, int JT = 2>  // XT: 0 = ci8 X, 1 = ci4 X
 __global__ __launch_bounds__(256) void beamform_mfma_kernel(
     long nn, long k, long nbatch, float alpha, const f2* __restrict__ w,
     long ldw, long w_b, const signed char* __restrict__ x, long ldx,
@@ -1309,7 +1309,7 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
     const int TW = 64 * JT;  // times per workgroup
     for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
         const f2* wb = w + batch * w_b;
-        const signed char* xb = x + 2 * (batch * x_b);
+        const signed char* xb = x + (XT ? 1 : 2) * (batch * x_b);
         f2* cb = c + batch * c_b;
         for (long j0 = (long)blockIdx.x * TW; j0 < nn;
              j0 += (long)gridDim.x * TW) {
@@ -1338,11 +1338,22 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                 }
                 for (int idx = tid; idx < TW * 8; idx += 256) {
                     int r = idx >> 3, q = idx & 7;  // time row, 8-k group
-                    const signed char* src =
-                        xb + 2 * ((j0 + r) * ldx + k0 + 8 * q);
                     int jt = r >> 4, rr = r & 15;
                     signed char buf[16];
-                    __builtin_memcpy(buf, src, 16);
+                    if (XT == 0) {  // ci8: 2 bytes per complex element
+                        __builtin_memcpy(
+                            buf, xb + 2 * ((j0 + r) * ldx + k0 + 8 * q),
+                            16);
+                    } else {  // ci4: re = HIGH nibble (linalg convention)
+                        unsigned char nib[8];
+                        __builtin_memcpy(
+                            nib, xb + (j0 + r) * ldx + k0 + 8 * q, 8);
+                        for (int e = 0; e < 8; ++e) {
+                            buf[2 * e] = (signed char)nib[e] >> 4;
+                            buf[2 * e + 1] =
+                                (signed char)(nib[e] << 4) >> 4;
+                        }
+                    }
                     // int8 values are exact in bf16, so the conversion is
                     // a float-bits truncation (low mantissa bits are 0);
                     // pack pairs into 32-bit LDS writes
@@ -1713,16 +1724,28 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
         // BIFROST_BEAM=valu.
         const char* beam_sel = getenv("BIFROST_BEAM");
         bool want_mfma = !(beam_sel && strcmp(beam_sel, "valu") == 0);
-        if (want_mfma && a_type == BF_DTYPE_CF32 && b_type == BF_DTYPE_CI8 &&
+        if (want_mfma && a_type == BF_DTYPE_CF32 &&
+            (b_type == BF_DTYPE_CI8 || b_type == BF_DTYPE_CI4) &&
             k % 64 == 0 && nn % 128 == 0 && m % 16 == 0 && nn > 0) {
             dim3 mgrid(cap_grid(nn / 128, 4096), cap_grid(nbatch, 65535));
+            bool x4 = b_type == BF_DTYPE_CI4;
             for (long i0 = 0; i0 < m;) {
                 long chunk = std::min<long>(64, m - i0);
 #define BEAM_MFMA_CASE(NBT)                                                   \
-    hipLaunchKernelGGL((beamform_mfma_kernel<NBT>), mgrid, dim3(256), 0,      \
-                       stream, nn, k, nbatch, (float)alpha, (const f2*)a,     \
-                       a_i, a_b, (const signed char*)b, b_j, b_b,             \
-                       (float)beta, (f2*)c, c_row, c_b, i0)
+    do {                                                                      \
+        if (x4)                                                               \
+            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, 1>), mgrid,         \
+                               dim3(256), 0, stream, nn, k, nbatch,           \
+                               (float)alpha, (const f2*)a, a_i, a_b,          \
+                               (const signed char*)b, b_j, b_b, (float)beta,  \
+                               (f2*)c, c_row, c_b, i0);                       \
+        else                                                                  \
+            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, 0>), mgrid,         \
+                               dim3(256), 0, stream, nn, k, nbatch,           \
+                               (float)alpha, (const f2*)a, a_i, a_b,          \
+                               (const signed char*)b, b_j, b_b, (float)beta,  \
+                               (f2*)c, c_row, c_b, i0);                       \
+    } while (0)
                 switch (chunk / 16) {
                 case 4: BEAM_MFMA_CASE(4); break;
                 case 3: BEAM_MFMA_CASE(3); break;

@@ -282,3 +282,26 @@ class TestBeamformerMFMA:
         b = np.asarray(c2.copy("system"))
         np.testing.assert_allclose(a, b, rtol=2e-4,
                                    atol=1e-3 * np.abs(b).max())
+
+    def test_mfma_ci4_input(self, linalg):
+        # config C5's actual input width: ci4 X through the MFMA path
+        np.random.seed(79)
+        ntime, nbeam, nstand, nchan = 128, 64, 32, 4
+        n = nstand * 2
+        re = np.random.randint(-7, 8, size=(ntime, nchan, n))
+        im = np.random.randint(-7, 8, size=(ntime, nchan, n))
+        packed = (((re & 0xF) << 4) | (im & 0xF)).astype(np.uint8)
+        x = (re + 1j * im).astype(np.complex64)
+        w = np.random.standard_normal(
+            (nbeam, nchan, n, 2)).astype(np.float32) \
+            .view(np.complex64).reshape(nbeam, nchan, n)
+        gold = np.matmul(w.transpose(1, 0, 2), x.transpose(1, 2, 0))
+        xb = bf.asarray(bf.ndarray(packed.view(bf.DataType.ci4)
+                                   .reshape(ntime, nchan, n)),
+                        space="cuda")
+        wb = bf.asarray(w, space="cuda")
+        c = bf.zeros_like(gold, space="cuda")
+        linalg.matmul(1, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0),
+                      0, c)
+        np.testing.assert_allclose(np.asarray(c.copy("system")), gold,
+                                   1e-4, 1e-3)
