@@ -1293,9 +1293,11 @@ typedef __bf16 bf16_t;
 typedef bf16_t v8bf __attribute__((ext_vector_type(8)));
 typedef float v4f __attribute__((ext_vector_type(4)));
 
-template <int NBT, int XT = 0, int JT = 2>  // XT: 0 = ci8 X, 1 = ci4 X
+// XT: 0 = ci8 X, 1 = ci4 X; WT: 0 = cf32 W (hi/lo bf16 RNE split),
+// 1 = ci16 W (exact high-byte/low-byte split)
+template <int NBT, int XT = 0, int WT = 0, int JT = 2>
 __global__ __launch_bounds__(256) void beamform_mfma_kernel(
-    long nn, long k, long nbatch, float alpha, const f2* __restrict__ w,
+    long nn, long k, long nbatch, float alpha, const void* __restrict__ w_,
     long ldw, long w_b, const signed char* __restrict__ x, long ldx,
     long x_b, float beta, f2* __restrict__ c, long c_row, long c_b,
     long i0) {
@@ -1308,7 +1310,8 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
     int row16 = lane & 15, kblk = lane >> 4;
     const int TW = 64 * JT;  // times per workgroup
     for (long batch = blockIdx.y; batch < nbatch; batch += gridDim.y) {
-        const f2* wb = w + batch * w_b;
+        const f2* wb = (const f2*)w_ + batch * w_b;        // WT == 0
+        const short* wb16 = (const short*)w_ + 2 * batch * w_b;  // WT == 1
         const signed char* xb = x + (XT ? 1 : 2) * (batch * x_b);
         f2* cb = c + batch * c_b;
         for (long j0 = (long)blockIdx.x * TW; j0 < nn;
@@ -1322,18 +1325,34 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
             for (long k0 = 0; k0 < k; k0 += 64) {
                 for (int idx = tid; idx < NBT * 16 * 16; idx += 256) {
                     int r = idx / 16, q = idx % 16;  // row, 4-k group
-                    const f2* src = wb + (i0 + r) * ldw + k0 + 4 * q;
                     int bt = r >> 4, rr = r & 15;
-                    for (int e = 0; e < 4; ++e) {
-                        f2 v = src[e];
-                        bf16_t hr = (bf16_t)v.x;
-                        bf16_t hi = (bf16_t)v.y;
-                        swp[bt][0][rr][4 * q + e] = hr;
-                        swp[bt][1][rr][4 * q + e] = hi;
-                        swp[bt][2][rr][4 * q + e] =
-                            (bf16_t)(v.x - (float)hr);
-                        swp[bt][3][rr][4 * q + e] =
-                            (bf16_t)(v.y - (float)hi);
+                    if (WT == 0) {
+                        const f2* src = wb + (i0 + r) * ldw + k0 + 4 * q;
+                        for (int e = 0; e < 4; ++e) {
+                            f2 v = src[e];
+                            bf16_t hr = (bf16_t)v.x;
+                            bf16_t hi = (bf16_t)v.y;
+                            swp[bt][0][rr][4 * q + e] = hr;
+                            swp[bt][1][rr][4 * q + e] = hi;
+                            swp[bt][2][rr][4 * q + e] =
+                                (bf16_t)(v.x - (float)hr);
+                            swp[bt][3][rr][4 * q + e] =
+                                (bf16_t)(v.y - (float)hi);
+                        }
+                    } else {  // ci16: exact byte split (both halves fit
+                              // bf16's 8-bit mantissa)
+                        const short* src =
+                            wb16 + 2 * ((i0 + r) * ldw + k0 + 4 * q);
+                        for (int e = 0; e < 4; ++e) {
+                            int vr = src[2 * e], vi = src[2 * e + 1];
+                            int hr = (vr >> 8) << 8, hi = (vi >> 8) << 8;
+                            swp[bt][0][rr][4 * q + e] = (bf16_t)(float)hr;
+                            swp[bt][1][rr][4 * q + e] = (bf16_t)(float)hi;
+                            swp[bt][2][rr][4 * q + e] =
+                                (bf16_t)(float)(vr - hr);
+                            swp[bt][3][rr][4 * q + e] =
+                                (bf16_t)(float)(vi - hi);
+                        }
                     }
                 }
                 for (int idx = tid; idx < TW * 8; idx += 256) {
@@ -1724,27 +1743,26 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
         // BIFROST_BEAM=valu.
         const char* beam_sel = getenv("BIFROST_BEAM");
         bool want_mfma = !(beam_sel && strcmp(beam_sel, "valu") == 0);
-        if (want_mfma && a_type == BF_DTYPE_CF32 &&
+        if (want_mfma &&
+            (a_type == BF_DTYPE_CF32 || a_type == BF_DTYPE_CI16) &&
             (b_type == BF_DTYPE_CI8 || b_type == BF_DTYPE_CI4) &&
             k % 64 == 0 && nn % 128 == 0 && m % 16 == 0 && nn > 0) {
             dim3 mgrid(cap_grid(nn / 128, 4096), cap_grid(nbatch, 65535));
             bool x4 = b_type == BF_DTYPE_CI4;
+            bool w16 = a_type == BF_DTYPE_CI16;
             for (long i0 = 0; i0 < m;) {
                 long chunk = std::min<long>(64, m - i0);
+#define BEAM_MFMA_ONE(NBT, XTV, WTV)                                          \
+    hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV>), mgrid,          \
+                       dim3(256), 0, stream, nn, k, nbatch, (float)alpha,     \
+                       a, a_i, a_b, (const signed char*)b, b_j, b_b,          \
+                       (float)beta, (f2*)c, c_row, c_b, i0)
 #define BEAM_MFMA_CASE(NBT)                                                   \
     do {                                                                      \
-        if (x4)                                                               \
-            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, 1>), mgrid,         \
-                               dim3(256), 0, stream, nn, k, nbatch,           \
-                               (float)alpha, (const f2*)a, a_i, a_b,          \
-                               (const signed char*)b, b_j, b_b, (float)beta,  \
-                               (f2*)c, c_row, c_b, i0);                       \
-        else                                                                  \
-            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, 0>), mgrid,         \
-                               dim3(256), 0, stream, nn, k, nbatch,           \
-                               (float)alpha, (const f2*)a, a_i, a_b,          \
-                               (const signed char*)b, b_j, b_b, (float)beta,  \
-                               (f2*)c, c_row, c_b, i0);                       \
+        if (x4 && w16) BEAM_MFMA_ONE(NBT, 1, 1);                              \
+        else if (x4) BEAM_MFMA_ONE(NBT, 1, 0);                                \
+        else if (w16) BEAM_MFMA_ONE(NBT, 0, 1);                               \
+        else BEAM_MFMA_ONE(NBT, 0, 0);                                        \
     } while (0)
                 switch (chunk / 16) {
                 case 4: BEAM_MFMA_CASE(4); break;
@@ -1753,6 +1771,7 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
                 default: BEAM_MFMA_CASE(1); chunk = 16; break;
                 }
 #undef BEAM_MFMA_CASE
+#undef BEAM_MFMA_ONE
                 BF_CHECK_HIP(hipGetLastError());
                 i0 += chunk;
             }

@@ -305,3 +305,29 @@ class TestBeamformerMFMA:
                       0, c)
         np.testing.assert_allclose(np.asarray(c.copy("system")), gold,
                                    1e-4, 1e-3)
+
+    def test_mfma_ci16_weights(self, linalg):
+        # ci16 weights: exact high/low-byte bf16 split through the MFMA
+        np.random.seed(80)
+        ntime, nbeam, nstand, nchan = 128, 32, 32, 4
+        n = nstand * 2
+        x8 = ((np.random.random((ntime, nchan, n, 2)) * 2 - 1)
+              * 127).astype(np.int8)
+        x = x8.astype(np.float32).view(np.complex64) \
+            .reshape(ntime, nchan, n)
+        w16 = np.random.randint(-3000, 3000,
+                                size=(nbeam, nchan, n, 2)).astype(np.int16)
+        w = (w16[..., 0].astype(np.float32) +
+             1j * w16[..., 1].astype(np.float32)).astype(np.complex64)
+        gold = np.matmul(w.transpose(1, 0, 2), x.transpose(1, 2, 0))
+        xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                   .reshape(ntime, nchan, n)),
+                        space="cuda")
+        wb = bf.asarray(bf.ndarray(w16.view(bf.DataType.ci16)
+                                   .reshape(nbeam, nchan, n)),
+                        space="cuda")
+        c = bf.zeros_like(gold, space="cuda")
+        linalg.matmul(1, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0),
+                      0, c)
+        np.testing.assert_allclose(np.asarray(c.copy("system")), gold,
+                                   1e-4, 1e-3 * np.abs(gold).max())
