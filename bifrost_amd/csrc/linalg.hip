@@ -937,15 +937,17 @@ void cherk_ci8_mfma_rs8_kernel(long n, long k, long nbatch, float alpha,
         for (int x = 0; x < 4; ++x)
             for (int y = 0; y < 4; ++y) acc[x][y] = v4i{};
 
-        const long slab_step = CHERK_BK * lda * 2;
-        const signed char* src0 = st_isJ
-            ? ab + (long)st_row * lda * 2 + j0 * 2 + 32 * st_q
-            : ab + (long)st_row * lda * 2 + i0 * 2 + 64 * st_q;
-        const signed char* load_next = src0;
+        // 32-bit staging cursor relative to `ab`: per-thread offsets fit
+        // u32 at hot-path sizes (k*lda*2 <= a few MB) and halve the VGPRs
+        // the 8-wave kernel spends on 64-bit address arithmetic
+        const unsigned slab_step32 = (unsigned)(CHERK_BK * lda * 2);
+        unsigned load_cur = st_isJ
+            ? (unsigned)((long)st_row * lda * 2 + j0 * 2 + 32 * st_q)
+            : (unsigned)((long)st_row * lda * 2 + i0 * 2 + 64 * st_q);
         v4i stg[4];
         auto load_slab = [&]() {
-            const signed char* p = load_next;
-            load_next += slab_step;
+            const signed char* p = ab + load_cur;
+            load_cur += slab_step32;
             if (st_isJ) {
                 const v4i* pv = (const v4i*)__builtin_assume_aligned(p, 16);
                 stg[0] = pv[0];
