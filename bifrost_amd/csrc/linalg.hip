@@ -1752,8 +1752,10 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
             dim3 mgrid(cap_grid(nn / 128, 4096), cap_grid(nbatch, 65535));
             bool x4 = b_type == BF_DTYPE_CI4;
             bool w16 = a_type == BF_DTYPE_CI16;
+            const char* ck = getenv("BIFROST_BEAM_CHUNK");
+            long max_chunk = ck ? atol(ck) : 64;
             for (long i0 = 0; i0 < m;) {
-                long chunk = std::min<long>(64, m - i0);
+                long chunk = std::min<long>(max_chunk, m - i0);
 #define BEAM_MFMA_ONE(NBT, XTV, WTV)                                          \
     hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV>), mgrid,          \
                        dim3(256), 0, stream, nn, k, nbatch, (float)alpha,     \
