@@ -61,3 +61,21 @@ samp = ntime * nchan
 flops = ntime*nchan*ks*nbeam*8
 print("beamform b64 s256 c512 t1024: %.3f ms, %.3f Gsamp/s, %.1f TFLOP/s"
       % (dt*1e3, samp/dt/1e9, flops/dt/1e12))
+
+# reduce: [8192, 4096] f32 time-scrunch by 16 (wave kernel path)
+A = bf.ndarray(shape=(8192, 4096), dtype="f32", space="cuda")
+B = bf.ndarray(shape=(8192, 256), dtype="f32", space="cuda")
+dt = timeit(lambda: bf.reduce(A, B, "mean"), n=10)
+gb = 8192*4096*4
+print("reduce f32 128MB axis-scrunch x16: %.3f ms, %.1f GB/s (read)"
+      % (dt*1e3, gb/dt/1e9))
+
+# fft: batched c2c 4096-point, 256 MB
+X = bf.ndarray(shape=(8192, 4096), dtype="cf32", space="cuda")
+Y = bf.ndarray(shape=(8192, 4096), dtype="cf32", space="cuda")
+f = bf.Fft()
+f.init(X, Y, axes=[1])
+dt = timeit(lambda: f.execute(X, Y), n=10)
+flops = 5 * 8192 * 4096 * 12  # 5 N log2 N
+print("fft c2c 8192x4096 batched: %.3f ms, %.1f GFLOP/s, %.1f GB/s (rw)"
+      % (dt*1e3, flops/dt/1e9, 2*8192*4096*8/dt/1e9))
