@@ -119,3 +119,32 @@ class TestGpuTranspose:
     def test_4d(self):
         self._run((4, 6, 8, 10), (3, 1, 2, 0), np.float32)
         self._run((4, 6, 8, 10), (0, 2, 1, 3), np.int16)
+
+
+class TestTranspose64Tile:
+    """The vectorized 64x64-tile path (4-byte elems, extents %64)."""
+
+    def test_2d_f32(self):
+        a = np.random.RandomState(0).standard_normal((256, 192)) \
+            .astype(np.float32)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.zeros((192, 256), dtype="f32", space="cuda")
+        bf.transpose(bg, ag, (1, 0))
+        np.testing.assert_array_equal(np.asarray(bg.copy("system")), a.T)
+
+    def test_3d_batched_i32(self):
+        a = np.random.RandomState(1).randint(
+            -10000, 10000, size=(6, 128, 64)).astype(np.int32)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.zeros((6, 64, 128), dtype="i32", space="cuda")
+        bf.transpose(bg, ag, (0, 2, 1))
+        np.testing.assert_array_equal(np.asarray(bg.copy("system")),
+                                      a.transpose(0, 2, 1))
+
+    def test_2d_odd_falls_back(self):
+        a = np.random.RandomState(2).standard_normal((130, 66)) \
+            .astype(np.float32)
+        ag = bf.asarray(a, space="cuda")
+        bg = bf.zeros((66, 130), dtype="f32", space="cuda")
+        bf.transpose(bg, ag, (1, 0))
+        np.testing.assert_array_equal(np.asarray(bg.copy("system")), a.T)
