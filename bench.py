@@ -207,11 +207,11 @@ def main():
     # --pmc passes (TCC_EA0_RDREQ_sum x 64 x 2 per the gfx950 FETCH_SIZE
     # calibration; see profiles/).  The default is the committed
     # measurement for the shipped sched-2 kernel at the default workload
-    # (profiles/round1_cherk.md: FETCH 5.20 GB/launch, L2 hit 73%);
+    # (profiles/round1_cherk.md: sched-5 FETCH 5.15 GB/launch, L2 hit 73%);
     # override with BIFROST_TRAFFIC_BYTES_PER_LAUNCH after re-profiling.
     traffic_env = os.environ.get("BIFROST_TRAFFIC_BYTES_PER_LAUNCH")
     if traffic_env is None and ntime == 4096 and nchan == 512 and N == 512:
-        traffic_env = "5.20e9"
+        traffic_env = "5.15e9"
     # The binding resource for the cherk kernel at n=512 is the i8 MFMA
     # pipe, not HBM: algorithmic intensity = 2n real-OPS per input byte =
     # 1024 OPS/B, above the machine balance (~7.9e15 int-OPS/s over 8e12
