@@ -1324,14 +1324,47 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                     accr[t][u] = v4f{0.f, 0.f, 0.f, 0.f};
                     acci[t][u] = v4f{0.f, 0.f, 0.f, 0.f};
                 }
-            for (long k0 = 0; k0 < k; k0 += 64) {
-                for (int idx = tid; idx < NBT * 16 * 16; idx += 256) {
-                    int r = idx / 16, q = idx % 16;  // row, 4-k group
+            // Register-prefetched staging: slab k0+64's global loads
+            // issue right after the barrier and fly across the whole MFMA
+            // burst (the 55 KB LDS footprint caps occupancy at 2
+            // waves/SIMD, so the prefetch registers are free).
+            signed char wpre[NBT][WT ? 16 : 32];
+            signed char xpre[2 * JT][XT ? 8 : 16];
+            auto fetch = [&](long k0) {
+                for (int it = 0; it < NBT; ++it) {
+                    int idx = tid + 256 * it;
+                    int r = idx / 16, q = idx % 16;
+                    if (WT == 0)
+                        __builtin_memcpy(
+                            wpre[it], wb + (i0 + r) * ldw + k0 + 4 * q,
+                            32);
+                    else
+                        __builtin_memcpy(
+                            wpre[it],
+                            wb16 + 2 * ((i0 + r) * ldw + k0 + 4 * q), 16);
+                }
+                for (int it = 0; it < 2 * JT; ++it) {
+                    int idx = tid + 256 * it;
+                    int r = idx >> 3, q = idx & 7;
+                    if (XT == 0)
+                        __builtin_memcpy(
+                            xpre[it],
+                            xb + 2 * ((j0 + r) * ldx + k0 + 8 * q), 16);
+                    else
+                        __builtin_memcpy(
+                            xpre[it], xb + (j0 + r) * ldx + k0 + 8 * q, 8);
+                }
+            };
+            auto commit = [&]() {
+                for (int it = 0; it < NBT; ++it) {
+                    int idx = tid + 256 * it;
+                    int r = idx / 16, q = idx % 16;
                     int bt = r >> 4, rr = r & 15;
                     if (WT == 0) {
-                        const f2* src = wb + (i0 + r) * ldw + k0 + 4 * q;
+                        f2 v4[4];
+                        __builtin_memcpy(v4, wpre[it], 32);
                         for (int e = 0; e < 4; ++e) {
-                            f2 v = src[e];
+                            f2 v = v4[e];
                             bf16_t hr = (bf16_t)v.x;
                             bf16_t hi = (bf16_t)v.y;
                             swp[bt][0][rr][4 * q + e] = hr;
@@ -1341,43 +1374,36 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                             swp[bt][3][rr][4 * q + e] =
                                 (bf16_t)(v.y - (float)hi);
                         }
-                    } else {  // ci16: exact byte split (both halves fit
-                              // bf16's 8-bit mantissa)
-                        const short* src =
-                            wb16 + 2 * ((i0 + r) * ldw + k0 + 4 * q);
+                    } else {
+                        short v8[8];
+                        __builtin_memcpy(v8, wpre[it], 16);
                         for (int e = 0; e < 4; ++e) {
-                            int vr = src[2 * e], vi = src[2 * e + 1];
-                            int hr = (vr >> 8) << 8, hi = (vi >> 8) << 8;
+                            int vr = v8[2 * e], vi = v8[2 * e + 1];
+                            int hr = (vr >> 8) << 8, hi2 = (vi >> 8) << 8;
                             swp[bt][0][rr][4 * q + e] = (bf16_t)(float)hr;
-                            swp[bt][1][rr][4 * q + e] = (bf16_t)(float)hi;
+                            swp[bt][1][rr][4 * q + e] =
+                                (bf16_t)(float)hi2;
                             swp[bt][2][rr][4 * q + e] =
                                 (bf16_t)(float)(vr - hr);
                             swp[bt][3][rr][4 * q + e] =
-                                (bf16_t)(float)(vi - hi);
+                                (bf16_t)(float)(vi - hi2);
                         }
                     }
                 }
-                for (int idx = tid; idx < TW * 8; idx += 256) {
-                    int r = idx >> 3, q = idx & 7;  // time row, 8-k group
+                for (int it = 0; it < 2 * JT; ++it) {
+                    int idx = tid + 256 * it;
+                    int r = idx >> 3, q = idx & 7;
                     int jt = r >> 4, rr = r & 15;
                     signed char buf[16];
-                    if (XT == 0) {  // ci8: 2 bytes per complex element
-                        __builtin_memcpy(
-                            buf, xb + 2 * ((j0 + r) * ldx + k0 + 8 * q),
-                            16);
-                    } else {  // ci4: re = HIGH nibble (linalg convention)
-                        unsigned char nib[8];
-                        __builtin_memcpy(
-                            nib, xb + (j0 + r) * ldx + k0 + 8 * q, 8);
+                    if (XT == 0) {
+                        __builtin_memcpy(buf, xpre[it], 16);
+                    } else {
                         for (int e = 0; e < 8; ++e) {
-                            buf[2 * e] = (signed char)nib[e] >> 4;
-                            buf[2 * e + 1] =
-                                (signed char)(nib[e] << 4) >> 4;
+                            unsigned char nb = (unsigned char)xpre[it][e];
+                            buf[2 * e] = (signed char)nb >> 4;
+                            buf[2 * e + 1] = (signed char)(nb << 4) >> 4;
                         }
                     }
-                    // int8 values are exact in bf16, so the conversion is
-                    // a float-bits truncation (low mantissa bits are 0);
-                    // pack pairs into 32-bit LDS writes
                     unsigned* pr = (unsigned*)&sxp[jt][0][rr][8 * q];
                     unsigned* pi = (unsigned*)&sxp[jt][1][rr][8 * q];
                     for (int e = 0; e < 4; ++e) {
@@ -1393,7 +1419,12 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                         pi[e] = i0b | (i1 << 16);
                     }
                 }
+            };
+            fetch(0);
+            for (long k0 = 0; k0 < k; k0 += 64) {
+                commit();
                 __syncthreads();
+                if (k0 + 64 < k) fetch(k0 + 64);
                 for (int kc = 0; kc < 2; ++kc) {
                     int kof = 32 * kc + 8 * kblk;
                     v8bf xr[JT], xi[JT], xni[JT];
