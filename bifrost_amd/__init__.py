@@ -6,9 +6,8 @@ alias `sys.modules['bifrost'] = bifrost_amd` for unmodified pipelines —
 see INTEGRATION.md).
 """
 
-__version__ = "0.1.0"
-
-from bifrost_amd import affinity, device, memory  # noqa: F401
+from bifrost_amd import affinity, core, device, memory  # noqa: F401
+from bifrost_amd.version import __version__  # noqa: F401
 from bifrost_amd import pipeline  # noqa: F401
 from bifrost_amd.pipeline import Pipeline, block_scope, get_default_pipeline  # noqa: F401
 from bifrost_amd.ring2 import Ring  # noqa: F401
