@@ -4,7 +4,7 @@ surface): stream a sequence to/from a simple on-disk format —
 offset>.dat` binary data files.
 
 Single-ringlet only (multi-ringlet rings are unsupported here,
-DESIGN.md §6); the reference's `<name>.bf.<offset>.<ringlet>.dat` form
+DESIGN.md out-of-scope list); the reference's `<name>.bf.<offset>.<ringlet>.dat` form
 is recognised but rejected on read.
 """
 

@@ -1,5 +1,5 @@
 """bf.map — apply a JIT'd function to named ndarrays (reference
-python/bifrost/map.py surface; elementwise subset)."""
+python/bifrost/map.py surface: elementwise AND axis-indexed forms)."""
 
 import ctypes
 
