@@ -79,3 +79,11 @@ dt = timeit(lambda: f.execute(X, Y), n=10)
 flops = 5 * 8192 * 4096 * 12  # 5 N log2 N
 print("fft c2c 8192x4096 batched: %.3f ms, %.1f GFLOP/s, %.1f GB/s (rw)"
       % (dt*1e3, flops/dt/1e9, 2*8192*4096*8/dt/1e9))
+
+# beamformer with ci4 input (config C5's 4-bit width)
+x4buf = np.random.randint(0, 256, size=(ntime, nchan, ks), dtype=np.uint8)
+x4 = bf.asarray(bf.ndarray(x4buf.view(bf.DataType.ci4)), space="cuda")
+x4v = x4.transpose(1, 2, 0)
+dt = timeit(lambda: la.matmul(1, wv, x4v, 0, bout), n=10)
+print("beamform ci4 b64 s256 c512 t1024: %.3f ms, %.3f Gsamp/s, %.1f TFLOP/s"
+      % (dt*1e3, samp/dt/1e9, flops/dt/1e12))
