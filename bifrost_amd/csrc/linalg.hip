@@ -1303,10 +1303,11 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
     long ldw, long w_b, const signed char* __restrict__ x, long ldx,
     long x_b, float beta, f2* __restrict__ c, long c_row, long c_b,
     long i0) {
-    // padded rows (72 = 64 + 8) keep v8bf reads 16B-aligned and spread
-    // the 16-row fragment reads over distinct banks
-    __shared__ bf16_t swp[NBT][4][16][72];   // [beam-tile][hr,hi,lr,li]
-    __shared__ bf16_t sxp[4 * JT][2][16][72];  // [time-tile][re,im]
+    // unpadded rows with a 16-B-chunk XOR swizzle (chunk ^ row&7, the
+    // cherk pattern): conflict-free fragment reads without the 8-element
+    // row padding — 49 KB total LDS = 3 workgroups/CU instead of 2
+    __shared__ bf16_t swp[NBT][4][16][64];   // [beam-tile][hr,hi,lr,li]
+    __shared__ bf16_t sxp[4 * JT][2][16][64];  // [time-tile][re,im]
     int tid = threadIdx.x;
     int wave = tid >> 6, lane = tid & 63;
     int row16 = lane & 15, kblk = lane >> 4;
@@ -1363,29 +1364,30 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                     if (WT == 0) {
                         f2 v4[4];
                         __builtin_memcpy(v4, wpre[it], 32);
+                        int wq = 8 * ((q >> 1) ^ (rr & 7)) + 4 * (q & 1);
                         for (int e = 0; e < 4; ++e) {
                             f2 v = v4[e];
                             bf16_t hr = (bf16_t)v.x;
                             bf16_t hi = (bf16_t)v.y;
-                            swp[bt][0][rr][4 * q + e] = hr;
-                            swp[bt][1][rr][4 * q + e] = hi;
-                            swp[bt][2][rr][4 * q + e] =
+                            swp[bt][0][rr][wq + e] = hr;
+                            swp[bt][1][rr][wq + e] = hi;
+                            swp[bt][2][rr][wq + e] =
                                 (bf16_t)(v.x - (float)hr);
-                            swp[bt][3][rr][4 * q + e] =
+                            swp[bt][3][rr][wq + e] =
                                 (bf16_t)(v.y - (float)hi);
                         }
                     } else {
                         short v8[8];
                         __builtin_memcpy(v8, wpre[it], 16);
+                        int wq = 8 * ((q >> 1) ^ (rr & 7)) + 4 * (q & 1);
                         for (int e = 0; e < 4; ++e) {
                             int vr = v8[2 * e], vi = v8[2 * e + 1];
                             int hr = (vr >> 8) << 8, hi2 = (vi >> 8) << 8;
-                            swp[bt][0][rr][4 * q + e] = (bf16_t)(float)hr;
-                            swp[bt][1][rr][4 * q + e] =
-                                (bf16_t)(float)hi2;
-                            swp[bt][2][rr][4 * q + e] =
+                            swp[bt][0][rr][wq + e] = (bf16_t)(float)hr;
+                            swp[bt][1][rr][wq + e] = (bf16_t)(float)hi2;
+                            swp[bt][2][rr][wq + e] =
                                 (bf16_t)(float)(vr - hr);
-                            swp[bt][3][rr][4 * q + e] =
+                            swp[bt][3][rr][wq + e] =
                                 (bf16_t)(float)(vi - hi2);
                         }
                     }
@@ -1404,8 +1406,9 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                             buf[2 * e + 1] = (signed char)(nb << 4) >> 4;
                         }
                     }
-                    unsigned* pr = (unsigned*)&sxp[jt][0][rr][8 * q];
-                    unsigned* pi = (unsigned*)&sxp[jt][1][rr][8 * q];
+                    int xq = 8 * (q ^ (rr & 7));
+                    unsigned* pr = (unsigned*)&sxp[jt][0][rr][xq];
+                    unsigned* pi = (unsigned*)&sxp[jt][1][rr][xq];
                     for (int e = 0; e < 4; ++e) {
                         unsigned r0 = __builtin_bit_cast(
                             unsigned, (float)buf[4 * e + 0]) >> 16;
@@ -1427,18 +1430,19 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                 if (k0 + 64 < k) fetch(k0 + 64);
                 for (int kc = 0; kc < 2; ++kc) {
                     int kof = 32 * kc + 8 * kblk;
+                    int sw = 8 * ((kof >> 3) ^ (row16 & 7));
                     v8bf xr[JT], xi[JT], xni[JT];
                     for (int u = 0; u < JT; ++u) {
                         int jt = wave * JT + u;
-                        xr[u] = *(const v8bf*)&sxp[jt][0][row16][kof];
-                        xi[u] = *(const v8bf*)&sxp[jt][1][row16][kof];
+                        xr[u] = *(const v8bf*)&sxp[jt][0][row16][sw];
+                        xi[u] = *(const v8bf*)&sxp[jt][1][row16][sw];
                         xni[u] = -xi[u];  // sign flip, packed ops
                     }
                     for (int bt = 0; bt < NBT; ++bt) {
-                        v8bf whr = *(const v8bf*)&swp[bt][0][row16][kof];
-                        v8bf whi = *(const v8bf*)&swp[bt][1][row16][kof];
-                        v8bf wlr = *(const v8bf*)&swp[bt][2][row16][kof];
-                        v8bf wli = *(const v8bf*)&swp[bt][3][row16][kof];
+                        v8bf whr = *(const v8bf*)&swp[bt][0][row16][sw];
+                        v8bf whi = *(const v8bf*)&swp[bt][1][row16][sw];
+                        v8bf wlr = *(const v8bf*)&swp[bt][2][row16][sw];
+                        v8bf wli = *(const v8bf*)&swp[bt][3][row16][sw];
                         for (int u = 0; u < JT; ++u) {
                             // Yr += (Whr+Wlr)·Xr − (Whi+Wli)·Xi
                             accr[bt][u] =

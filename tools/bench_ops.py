@@ -76,9 +76,9 @@ Y = bf.ndarray(shape=(8192, 4096), dtype="cf32", space="cuda")
 f = bf.Fft()
 f.init(X, Y, axes=[1])
 dt = timeit(lambda: f.execute(X, Y), n=10)
-flops = 5 * 8192 * 4096 * 12  # 5 N log2 N
+fft_flops = 5 * 8192 * 4096 * 12  # 5 N log2 N
 print("fft c2c 8192x4096 batched: %.3f ms, %.1f GFLOP/s, %.1f GB/s (rw)"
-      % (dt*1e3, flops/dt/1e9, 2*8192*4096*8/dt/1e9))
+      % (dt*1e3, fft_flops/dt/1e9, 2*8192*4096*8/dt/1e9))
 
 # beamformer with ci4 input (config C5's 4-bit width)
 x4buf = np.random.randint(0, 256, size=(ntime, nchan, ks), dtype=np.uint8)
