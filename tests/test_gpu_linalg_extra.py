@@ -98,3 +98,39 @@ def test_quantize_real_dtypes(dtype):
     want = oracle.quantize(data, dtype, scale=1.0)
     got = np.asarray(o.copy("system"))
     np.testing.assert_array_equal(got.view(want.dtype).reshape(-1), want)
+
+
+class TestCherkVariantSelectors:
+    """The documented BIFROST_CHERK/_SCHED selectors all stay correct."""
+
+    @pytest.mark.parametrize("sel,env", [
+        ("rs", {}),
+        ("rs", {"BIFROST_CHERK_SCHED": "0"}),
+        ("rs", {"BIFROST_CHERK_SCHED": "1"}),
+        ("rs", {"BIFROST_CHERK_SCHED": "2"}),
+        ("rs", {"BIFROST_CHERK_SCHED": "3"}),
+        ("rs8", {}),
+        ("wave", {}),
+        ("coop", {}),
+        ("pipe", {}),
+    ])
+    def test_variant(self, linalg, monkeypatch, sel, env):
+        monkeypatch.setenv("BIFROST_CHERK", sel)
+        for k, v in env.items():
+            monkeypatch.setenv(k, v)
+        np.random.seed(4321)
+        ntime, n, nchan = 256, 128, 3
+        x8 = ((np.random.random((ntime, nchan, n, 2)) * 2 - 1) * 127) \
+            .astype(np.int8)
+        x = x8.astype(np.float32).view(np.complex64).reshape(ntime, nchan, n)
+        xs = x.transpose(1, 0, 2)
+        gold = np.matmul(np.conj(xs.transpose(0, 2, 1)), xs)
+        for c in range(nchan):
+            gold[c][np.triu_indices(n, 1)] = 0
+        xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                   .reshape(ntime, nchan, n)),
+                        space="cuda").transpose(1, 0, 2)
+        cb = bf.zeros((nchan, n, n), dtype="cf32", space="cuda")
+        linalg.matmul(1, None, xb, 0, cb)
+        np.testing.assert_allclose(np.asarray(cb.copy("system")), gold,
+                                   1e-4, 1e-5 * max(1, np.abs(gold).max()))
