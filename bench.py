@@ -206,7 +206,7 @@ def main():
     # Measured HBM/fabric traffic per launch comes from separate rocprofv3
     # --pmc passes (TCC_EA0_RDREQ_sum x 64 x 2 per the gfx950 FETCH_SIZE
     # calibration; see profiles/).  The default is the committed
-    # measurement for the shipped sched-2 kernel at the default workload
+    # measurement for the shipped sched-5 kernel at the default workload
     # (profiles/round1_cherk.md: sched-5 FETCH 5.15 GB/launch, L2 hit 73%);
     # override with BIFROST_TRAFFIC_BYTES_PER_LAUNCH after re-profiling.
     traffic_env = os.environ.get("BIFROST_TRAFFIC_BYTES_PER_LAUNCH")

@@ -222,7 +222,9 @@ __global__ __launch_bounds__(256) void herk_generic_kernel(long n, long k, long 
 //
 // Workgroup: 256 threads / 4 waves; 64x64 complex output tile; each wave a
 // 32x32 quadrant = 4x4 byte-tiles of 16x16 = 16 MFMA per K-slab of 64.
-// Accumulators: 16 x v4i = 64 AGPRs.  Epilogue pairs columns across even/
+// Accumulators: 16 x v4i = 64 regs (VGPRs: the gfx950 unified file means
+// the backend keeps them there, no AGPR split).  Epilogue pairs columns
+// across even/
 // odd lanes with one shfl_xor.  Dispatch requires n, lda, batch stride
 // even (the reference's own routing conditions); edge tiles zero-pad.
 
