@@ -10,6 +10,7 @@ from bifrost_amd import affinity, core, device, memory  # noqa: F401
 from bifrost_amd.version import __version__  # noqa: F401
 from bifrost_amd import pipeline  # noqa: F401
 from bifrost_amd.pipeline import Pipeline, block_scope, get_default_pipeline  # noqa: F401
+from bifrost_amd import ring  # noqa: F401  (classic byte-span API)
 from bifrost_amd.ring2 import Ring  # noqa: F401
 from bifrost_amd import blocks  # noqa: F401
 from bifrost_amd import views  # noqa: F401
