@@ -88,3 +88,53 @@ def test_classic_sequence_generator_across_sequences():
     t.join()
     assert names == ["seq0", "seq1", "seq2"]
     assert payloads == [0, 1, 2]
+
+
+def test_classic_late_resize_preserves_data():
+    """Reference test_resizing.py behaviour at the ring level: growing the
+    ring AFTER a read sequence is open (and data written) must preserve the
+    live bytes and keep reads working."""
+    ring = Ring(space="system", name="classic-lateresize")
+    gulp = 32
+    ring.resize(gulp, 2 * gulp)
+
+    def writer():
+        with ring.begin_writing() as ow:
+            with ow.begin_sequence(name="s") as oseq:
+                for i in range(2):
+                    with oseq.reserve(gulp) as wspan:
+                        wspan.data_view()[0, :] = i
+
+    t = threading.Thread(target=writer)
+    t.start()
+    with ring.open_earliest_sequence(guarantee=True) as iseq:
+        spans = iseq.read(gulp)
+        first = next(spans).data.tobytes()
+        # late resize mid-sequence (bigger gulp AND bigger total)
+        ring.resize(4 * gulp, 16 * gulp)
+        rest = [s.data.tobytes() for s in spans]
+    t.join()
+    assert first == bytes([0] * gulp)
+    assert rest == [bytes([1] * gulp)]
+
+
+def test_classic_gulp_larger_than_ring_grows_capacity():
+    """Reference TestLargeGulpSize: requesting a contiguous span larger
+    than the ring's total size must grow the ring, not fault."""
+    ring = Ring(space="system", name="classic-largegulp")
+    ring.resize(12)            # tiny: total 48 bytes
+    ring.resize(1024, 1024)    # "gulp" bigger than the old ring
+    payload = np.resize(np.arange(256, dtype=np.uint8), 1024)
+
+    def writer():
+        with ring.begin_writing() as ow:
+            with ow.begin_sequence(name="big") as oseq:
+                with oseq.reserve(1024) as wspan:
+                    wspan.data_view()[0, :] = payload
+
+    t = threading.Thread(target=writer)
+    t.start()
+    with ring.open_earliest_sequence(guarantee=True) as iseq:
+        got = [s.data.tobytes() for s in iseq.read(1024)]
+    t.join()
+    assert got == [payload.tobytes()]
