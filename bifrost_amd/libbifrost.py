@@ -206,6 +206,7 @@ _proto("bfAffinityGetCore", BFstatus, c_int_p)
 _proto("bfAffinitySetOpenMPCores", BFstatus, BFsize, c_int_p)
 
 # ring
+_proto("bfTestSuite", ctypes.c_int)
 _proto("bfRingCreate", BFstatus, ctypes.POINTER(_bf.BFring), ctypes.c_char_p,
        BFspace)
 _proto("bfRingDestroy", BFstatus, _bf.BFring)

@@ -6,6 +6,7 @@ hand-written FFI layer."""
 from bifrost_amd.libbifrost import _bf, BFarray as struct_BFarray_  # noqa: F401
 
 BF_CUDA_ENABLED = _bf.BF_CUDA_ENABLED
+bfTestSuite = _bf.bfTestSuite  # C self-test entry (reference test_library.py)
 BF_FLOAT128_ENABLED = _bf.BF_FLOAT128_ENABLED
 BF_DEBUG_ENABLED = _bf.BF_DEBUG_ENABLED
 BF_TRACE_ENABLED = _bf.BF_TRACE_ENABLED
