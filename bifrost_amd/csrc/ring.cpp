@@ -60,6 +60,13 @@ struct BFring_impl {
     BFoffset reserve_head = 0;  // frontier of reserved (uncommitted) bytes
     BFoffset head = 0;          // frontier of committed bytes
 
+    // Open-span accounting so resize can wait for quiescence instead of
+    // reallocating under a live span pointer (reference
+    // ring_impl.cpp:73-80,515-524 RingReallocLock semantics).
+    BFsize nwrite_open = 0;
+    BFsize nread_open = 0;
+    BFsize nrealloc_pending = 0;
+
     std::deque<std::shared_ptr<Sequence>> sequences;
     uint64_t next_seq_id = 0;
 
@@ -160,6 +167,22 @@ BFstatus bfRingResize(BFring ring, BFsize contiguous_bytes,
         BFsize new_cap = std::max(ring->capacity,
                                   std::max(capacity_bytes, new_ghost));
         if (new_ghost == ring->ghost && new_cap == ring->capacity) return BF_STATUS_SUCCESS;
+        // Wait until no span is open: live spans hold raw pointers into the
+        // old buffer, so reallocating under them is a use-after-free.  New
+        // reserve/acquire calls block while nrealloc_pending > 0.
+        ++ring->nrealloc_pending;
+        ring->cv.wait(lk, [&] {
+            return ring->nwrite_open == 0 && ring->nread_open == 0;
+        });
+        --ring->nrealloc_pending;
+        // Re-check: another resize may have satisfied the request meanwhile.
+        new_ghost = std::max(ring->ghost, contiguous_bytes);
+        new_cap = std::max(ring->capacity,
+                           std::max(capacity_bytes, new_ghost));
+        if (new_ghost == ring->ghost && new_cap == ring->capacity) {
+            ring->cv.notify_all();
+            return BF_STATUS_SUCCESS;
+        }
         char* new_buf = nullptr;
         BF_THROW_IF(bfMalloc((void**)&new_buf, new_cap + new_ghost,
                              ring->space) != BF_STATUS_SUCCESS,
@@ -464,6 +487,12 @@ BFstatus bfRingSpanReserve(BFwspan* span, BFring ring,
     // Block while committing would overwrite data a guaranteed reader still
     // needs.  Non-guaranteed data is overwritten freely (readers detect it).
     for (;;) {
+        if (ring->nrealloc_pending) {
+            // A resize is waiting for quiescence; don't open new spans.
+            if (nonblocking) return BF_STATUS_WOULD_BLOCK;
+            ring->cv.wait(lk);
+            continue;
+        }
         BFoffset gt = ring->guarded_tail();
         if (ring->reserve_head + size <= gt + ring->capacity) break;
         if (ring->writing_ended) return BF_STATUS_INVALID_STATE;
@@ -480,6 +509,7 @@ BFstatus bfRingSpanReserve(BFwspan* span, BFring ring,
     // (partial commits elsewhere could otherwise be clobbered by flush).
     BFstatus st = ring->refresh_ghost(ws->begin, size);
     if (st != BF_STATUS_SUCCESS) { delete ws; return st; }
+    ++ring->nwrite_open;
     *span = ws;
     return BF_STATUS_SUCCESS;
 }
@@ -491,9 +521,15 @@ BFstatus bfRingSpanCommit(BFwspan span, BFsize size) {
     {
         std::lock_guard<std::mutex> lk(ring->mutex);
         BFstatus st = ring->flush_ghost(span->begin, size);
-        if (st != BF_STATUS_SUCCESS) { delete span; return st; }
+        if (st != BF_STATUS_SUCCESS) {
+            --ring->nwrite_open;
+            ring->cv.notify_all();
+            delete span;
+            return st;
+        }
         ring->head = span->begin + size;
         ring->reserve_head = ring->head;
+        --ring->nwrite_open;
         ring->cv.notify_all();
     }
     delete span;
@@ -526,6 +562,11 @@ BFstatus bfRingSpanAcquire(BFrspan* span, BFrsequence sequence,
         ring->cv.notify_all();
     }
     for (;;) {
+        if (ring->nrealloc_pending) {
+            // A resize is waiting for quiescence; don't open new spans.
+            ring->cv.wait(lk);
+            continue;
+        }
         if (seq->ended && req_begin >= seq->end) return BF_STATUS_END_OF_DATA;
         if (ring->writing_ended && !seq->ended && req_begin >= ring->head)
             return BF_STATUS_END_OF_DATA;
@@ -555,6 +596,7 @@ BFstatus bfRingSpanAcquire(BFrspan* span, BFrsequence sequence,
                 if (st != BF_STATUS_SUCCESS) return st;
             }
             sequence->guard = begin;
+            ++ring->nread_open;
             auto* rs = new BFrspan_impl();
             rs->is_write = false;
             rs->ring = ring;
@@ -577,6 +619,7 @@ BFstatus bfRingSpanRelease(BFrspan span) {
         std::lock_guard<std::mutex> lk(ring->mutex);
         // Advance the reader's guard past this span (sequential-gulp model).
         rseq->guard = std::max(rseq->guard, span->begin + span->size);
+        --ring->nread_open;
         ring->cv.notify_all();
     }
     delete span;

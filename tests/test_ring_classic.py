@@ -91,9 +91,11 @@ def test_classic_sequence_generator_across_sequences():
 
 
 def test_classic_late_resize_preserves_data():
-    """Reference test_resizing.py behaviour at the ring level: growing the
-    ring AFTER a read sequence is open (and data written) must preserve the
-    live bytes and keep reads working."""
+    """Reference test_resizing.py behaviour: growing the ring AFTER a read
+    sequence is open (the ModResizeAsciiBlock pattern — resize between
+    opening the sequence and acquiring spans) must preserve live data.
+    The resize waits for span quiescence (no open spans), so it must be
+    called while no span is held."""
     ring = Ring(space="system", name="classic-lateresize")
     gulp = 32
     ring.resize(gulp, 2 * gulp)
@@ -108,14 +110,34 @@ def test_classic_late_resize_preserves_data():
     t = threading.Thread(target=writer)
     t.start()
     with ring.open_earliest_sequence(guarantee=True) as iseq:
-        spans = iseq.read(gulp)
-        first = next(spans).data.tobytes()
-        # late resize mid-sequence (bigger gulp AND bigger total)
+        # late resize: after the sequence is open, before any span is held
         ring.resize(4 * gulp, 16 * gulp)
-        rest = [s.data.tobytes() for s in spans]
+        got = [s.data.tobytes() for s in iseq.read(gulp)]
     t.join()
-    assert first == bytes([0] * gulp)
-    assert rest == [bytes([1] * gulp)]
+    assert got == [bytes([0] * gulp), bytes([1] * gulp)]
+
+
+def test_classic_resize_preserves_committed_data():
+    """Growing the ring after data is fully written re-places the live
+    bytes at their new positions (no torn reads)."""
+    ring = Ring(space="system", name="classic-resize-keep")
+    gulp = 48
+    ring.resize(gulp, 4 * gulp)
+
+    def writer():
+        with ring.begin_writing() as ow:
+            with ow.begin_sequence(name="s") as oseq:
+                for i in range(4):
+                    with oseq.reserve(gulp) as wspan:
+                        wspan.data_view()[0, :] = 10 + i
+
+    t = threading.Thread(target=writer)
+    t.start()
+    t.join()
+    ring.resize(2 * gulp, 16 * gulp)   # grow: ghost AND capacity
+    with ring.open_earliest_sequence(guarantee=True) as iseq:
+        got = [s.data.tobytes() for s in iseq.read(gulp)]
+    assert got == [bytes([10 + i] * gulp) for i in range(4)]
 
 
 def test_classic_gulp_larger_than_ring_grows_capacity():
