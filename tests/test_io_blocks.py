@@ -58,6 +58,33 @@ class TestSerialize:
         data = bf.ndarray(raw.view(bf.DataType.ci8)).reshape(16, 4)
         self._roundtrip(tmp_path, data)
 
+    def test_serialize_unnamed_uses_time_tag(self, tmp_path):
+        # An unnamed sequence serializes under its zero-padded time_tag
+        # (reference test_serialize.py:108-126 / blocks/serialize.py).
+        class UnnamedSource(NumpySourceBlock):
+            def on_sequence(self, reader, sourcename):
+                hdrs = super(UnnamedSource, self).on_sequence(reader,
+                                                              sourcename)
+                for h in hdrs:
+                    h["name"] = ""
+                    h["time_tag"] = 1234
+                return hdrs
+
+        data = np.arange(8 * 3, dtype=np.float32).reshape(8, 3)
+        with bf.Pipeline() as pipe:
+            src = UnnamedSource([data], gulp_nframe=4)
+            bf.blocks.serialize(src, path=str(tmp_path))
+            pipe.run()
+        base = str(tmp_path / ("%020i.bf" % 1234))
+        assert glob.glob(base + ".json"), "expected time-tag-named header"
+
+        out = []
+        with bf.Pipeline() as pipe:
+            src2 = bf.blocks.deserialize([base], gulp_nframe=4)
+            CollectBlock(src2, out)
+            pipe.run()
+        np.testing.assert_array_equal(np.concatenate(out, axis=0), data)
+
 
 # ---------------------------------------------------------------------------
 # binary read / write
