@@ -141,6 +141,34 @@ class Pipeline(BlockScope):
         for b in self.blocks:
             b.shutdown()
 
+    def dot_graph(self):
+        """DOT source for the block/ring graph (reference
+        pipeline.py:163-201 dot_graph; emitted as plain DOT text rather
+        than a graphviz.Digraph so no external package is needed —
+        render with `dot -Tsvg`)."""
+        space_colors = {"system": "orange", "cuda": "limegreen",
+                        "cuda_host": "deepskyblue"}
+        lines = ["digraph \"cluster_%s\" {" % self.name]
+        rings = {}
+        for block in self.blocks:
+            label = block.name.split("/", 1)[-1]
+            fill = ("lightsteelblue"
+                    if block.__class__.__name__ == "CopyBlock" else "white")
+            lines.append('  "%s" [label="%s", shape=box, style=filled, '
+                         'fillcolor=%s];' % (block.name, label, fill))
+            for oring in block.orings:
+                rings[oring.name] = oring
+                lines.append('  "%s" -> "%s";' % (block.name, oring.name))
+            for iring in block.irings:
+                rings[iring.name] = iring
+                lines.append('  "%s" -> "%s";' % (iring.name, block.name))
+        for name, ring in sorted(rings.items()):
+            color = space_colors.get(ring.space, "white")
+            lines.append('  "%s" [shape=ellipse, style=filled, '
+                         'fillcolor=%s];' % (name, color))
+        lines.append("}")
+        return "\n".join(lines)
+
     def __enter__(self):
         self._prev_default = get_default_pipeline()
         self.as_default()

@@ -167,3 +167,20 @@ def test_pipeline_propagates_block_errors():
             src = NumpySourceBlock([data], gulp_nframe=2)
             BoomBlock(src)
             pipe.run()
+
+
+def test_pipeline_dot_graph():
+    # Reference pipeline.py:163-201: blocks as boxes, rings as colored
+    # ellipses, edges block->oring and iring->block.
+    data = np.ones((8, 3), dtype=np.float32)
+    out = []
+    with Pipeline() as pipe:
+        src = NumpySourceBlock([data], gulp_nframe=4)
+        cpy = bf.blocks.copy(src)
+        CollectBlock(cpy, out)
+        dot = pipe.dot_graph()
+    assert dot.startswith("digraph")
+    assert "shape=box" in dot and "shape=ellipse" in dot
+    assert "fillcolor=lightsteelblue" in dot     # CopyBlock color
+    assert "fillcolor=orange" in dot             # system-space ring
+    assert dot.count("->") >= 4
