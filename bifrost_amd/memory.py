@@ -6,7 +6,8 @@ import ctypes
 from bifrost_amd.libbifrost import _bf, _check, _string2space, _space2string
 
 __all__ = ["raw_malloc", "raw_free", "raw_get_space", "space_accessible",
-           "alignment", "memcpy", "memset"]
+           "alignment", "memcpy", "memcpy2D", "memset", "memset2D",
+           "raw_memcpy", "raw_memset"]
 
 _ACCESSIBILITY = {
     "system": {"system"},
@@ -46,10 +47,56 @@ def alignment():
     return int(_bf.bfGetAlignment())
 
 
-def memcpy(dst_ptr, dst_space, src_ptr, src_space, count):
+def raw_memcpy(dst_ptr, dst_space, src_ptr, src_space, count):
     _check(_bf.bfMemcpy(dst_ptr, _string2space(str(dst_space)), src_ptr,
                         _string2space(str(src_space)), count))
 
 
-def memset(ptr, space, value, count):
+def raw_memset(ptr, space, value, count):
     _check(_bf.bfMemset(ptr, _string2space(str(space)), value, count))
+
+
+def _get_space(arr):
+    """Space of a numpy/bifrost array ('system' for plain numpy —
+    reference memory.py:_get_space)."""
+    try:
+        return arr.flags["SPACE"]
+    except (AttributeError, KeyError):
+        return "system"
+
+
+# Array-based forms, matching the reference memory.py call surface
+# (memcpy(dst, src) etc. over numpy or bifrost ndarrays).
+
+def memcpy(dst, src):
+    assert dst.flags["C_CONTIGUOUS"]
+    assert src.shape == dst.shape
+    _check(_bf.bfMemcpy(dst.ctypes.data, _string2space(_get_space(dst)),
+                        src.ctypes.data, _string2space(_get_space(src)),
+                        dst.nbytes))
+    return dst
+
+
+def memcpy2D(dst, src):
+    assert len(dst.shape) == 2
+    assert src.shape == dst.shape
+    width_bytes = dst.shape[1] * dst.dtype.itemsize
+    _check(_bf.bfMemcpy2D(dst.ctypes.data, dst.strides[0],
+                          _string2space(_get_space(dst)),
+                          src.ctypes.data, src.strides[0],
+                          _string2space(_get_space(src)),
+                          width_bytes, dst.shape[0]))
+
+
+def memset(dst, val=0):
+    assert dst.flags["C_CONTIGUOUS"]
+    _check(_bf.bfMemset(dst.ctypes.data, _string2space(_get_space(dst)),
+                        val, dst.nbytes))
+
+
+def memset2D(dst, val=0):
+    assert len(dst.shape) == 2
+    width_bytes = dst.shape[1] * dst.dtype.itemsize
+    _check(_bf.bfMemset2D(dst.ctypes.data, dst.strides[0],
+                          _string2space(_get_space(dst)), val,
+                          width_bytes, dst.shape[0]))
