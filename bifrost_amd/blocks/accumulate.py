@@ -1,23 +1,31 @@
 """AccumulateBlock (reference blocks/accumulate.py surface): sum frames
-over an integration window."""
+over an integration window, optionally converting dtype (`dtype` kwarg,
+e.g. accumulate ci8 into cf32).  System-space rings accumulate with
+numpy; device rings use bf.map with the reference's
+`b = beta*b + (b_type)a` kernel (reference accumulate.py:AccumulateBlock)."""
 
 from copy import deepcopy
+import importlib
 
 import numpy as np
 
+from bifrost_amd.DataType import DataType
 from bifrost_amd.pipeline import TransformBlock
+
+_map = importlib.import_module("bifrost_amd.map")
 
 __all__ = ["AccumulateBlock", "accumulate"]
 
 
 class AccumulateBlock(TransformBlock):
-    def __init__(self, iring, nframe_to_accumulate, *args, **kwargs):
-        super(AccumulateBlock, self).__init__(iring, *args, **kwargs)
-        self.nframe_to_accumulate = nframe_to_accumulate
+    def __init__(self, iring, nframe, dtype=None, *args, **kwargs):
         kwargs.setdefault("gulp_nframe", 1)
+        super(AccumulateBlock, self).__init__(iring, *args, **kwargs)
+        self.nframe = nframe
+        self.dtype = dtype
 
     def define_valid_input_spaces(self):
-        return ("system",)
+        return "any"
 
     def define_output_nframes(self, input_nframe):
         return 1
@@ -25,30 +33,53 @@ class AccumulateBlock(TransformBlock):
     def on_sequence(self, iseq):
         self.nframe_accumulated = 0
         ohdr = deepcopy(iseq.header)
-        if "scales" in ohdr["_tensor"] and ohdr["_tensor"]["scales"][0]:
-            ohdr["_tensor"]["scales"][0][1] *= self.nframe_to_accumulate
+        otensor = ohdr["_tensor"]
+        if "scales" in otensor:
+            frame_axis = otensor["shape"].index(-1)
+            if otensor["scales"][frame_axis]:
+                otensor["scales"][frame_axis][1] *= self.nframe
+        if self.dtype is not None:
+            otensor["dtype"] = str(DataType(self.dtype))
         return ohdr
 
     def on_data(self, ispan, ospan):
-        idata = np.asarray(ispan.data)
-        odata = np.asarray(ospan.data)
-        summed = idata.sum(axis=0, keepdims=True)
-        if self.nframe_accumulated == 0:
-            odata[...] = summed
+        first = self.nframe_accumulated == 0
+        if getattr(ispan.data, "bf", None) is not None and \
+                ispan.data.bf.space == "cuda":
+            # the reference's kernel, split by beta to avoid a scalar arg
+            func = "b = (b_type)a" if first else "b += (b_type)a"
+            _map.map(func, {"a": ispan.data, "b": ospan.data})
         else:
-            odata[...] += summed
+            idata = np.asarray(ispan.data)
+            odata = np.asarray(ospan.data)
+            if idata.dtype.names:
+                # complex-integer input: only a complex-float output can
+                # hold the running sum on the CPU path
+                if odata.dtype.names:
+                    raise NotImplementedError(
+                        "system-space accumulate of complex-integer data "
+                        "requires dtype='cf32'/'cf64'")
+                flt = idata.view(idata.dtype[0]).astype(np.float32)
+                idata = flt.view(np.complex64)
+            summed = idata.sum(axis=0, keepdims=True)
+            if first:
+                odata[...] = summed
+            else:
+                odata[...] += summed
         self.nframe_accumulated += ispan.nframe
-        assert self.nframe_accumulated <= self.nframe_to_accumulate
-        if self.nframe_accumulated == self.nframe_to_accumulate:
+        assert self.nframe_accumulated <= self.nframe
+        if self.nframe_accumulated == self.nframe:
             self.nframe_accumulated = 0
             return 1
         return 0
 
 
-def accumulate(iring, nframe_to_accumulate, *args, **kwargs):
-    """Accumulate (sum) frames over a window.
+def accumulate(iring, nframe, dtype=None, *args, **kwargs):
+    """Accumulate (sum) `nframe` frames per output frame.
 
-    Input:  [...], any numeric dtype, space = system
-    Output: [...], same dtype, 1 frame per window
+    dtype: output datatype (default: same as input).
+
+    Input:  [..., 'time', ...], dtype = any, space = system or CUDA
+    Output: [..., 'time'/nframe, ...], 1 frame per window
     """
-    return AccumulateBlock(iring, nframe_to_accumulate, *args, **kwargs)
+    return AccumulateBlock(iring, nframe, dtype, *args, **kwargs)

@@ -184,3 +184,23 @@ def test_pipeline_dot_graph():
     assert "fillcolor=lightsteelblue" in dot     # CopyBlock color
     assert "fillcolor=orange" in dot             # system-space ring
     assert dot.count("->") >= 4
+
+
+def test_pipeline_accumulate_block_dtype_upconvert():
+    # reference accumulate.py: dtype kwarg converts the output datatype
+    # (here ci8 frames summed into a cf32 accumulator on the CPU path)
+    raw = np.zeros((8, 4), dtype=[("re", np.int8), ("im", np.int8)])
+    raw["re"] = np.arange(32).reshape(8, 4) % 5
+    raw["im"] = 1
+    out = []
+    with Pipeline() as pipe:
+        src = NumpySourceBlock([raw], gulp_nframe=1)
+        acc = bf.blocks.accumulate(src, 4, dtype="cf32", gulp_nframe=1)
+        CollectBlock(acc, out)
+        pipe.run()
+    got = np.concatenate(out, axis=0)
+    assert got.dtype == np.complex64
+    want = (raw["re"].astype(np.float32)
+            + 1j * raw["im"].astype(np.float32))
+    np.testing.assert_allclose(got[0], want[0:4].sum(axis=0))
+    np.testing.assert_allclose(got[1], want[4:8].sum(axis=0))
