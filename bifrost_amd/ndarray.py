@@ -300,6 +300,12 @@ class ndarray(np.ndarray):
             return self
         return self.copy(space="system")
 
+    def tofile(self, fid, sep="", format="%s"):
+        # device arrays stage through a system copy first (reference
+        # ndarray.py tofile override)
+        return super(ndarray,
+                     self._system_accessible_copy()).tofile(fid, sep, format)
+
     def __repr__(self):
         return super(ndarray, self._system_accessible_copy()).__repr__()
 

@@ -110,3 +110,14 @@ def test_BFarray_roundtrip():
     aa = a.as_BFarray()
     b = bf.ndarray(aa)
     np.testing.assert_equal(np.asarray(a), np.asarray(b))
+
+
+def test_tofile_system(tmp_path):
+    import os
+    a = bf.ndarray(np.arange(16, dtype=np.float32))
+    path = str(tmp_path / "a.dat")
+    with open(path, "wb") as f:
+        a.tofile(f)
+    assert os.path.getsize(path) == 64
+    np.testing.assert_array_equal(np.fromfile(path, dtype=np.float32),
+                                  np.arange(16, dtype=np.float32))
