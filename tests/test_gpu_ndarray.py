@@ -167,3 +167,14 @@ def test_torch_buffer_interop():
     got = t.cpu().numpy()
     np.testing.assert_array_equal(got[:, 0], np.full(16, 7.0))
     np.testing.assert_array_equal(got[:, 1], np.full(16, -3.0))
+
+
+@pytest.mark.gpu
+def test_device_tofile(tmp_path):
+    host = np.arange(1024, dtype=np.float32).reshape(32, 32)
+    dev = bf.asarray(bf.ndarray(host), space="cuda")
+    path = str(tmp_path / "dev.dat")
+    with open(path, "wb") as f:
+        dev.tofile(f)
+    back = np.fromfile(path, dtype=np.float32).reshape(32, 32)
+    np.testing.assert_array_equal(back, host)
