@@ -158,3 +158,49 @@ def test_correlator_oracle_time_split_equals_whole():
     summed = sum(np.matmul(H(x[:, lo:hi]), x[:, lo:hi])
                  for lo, hi in ((0, 8), (8, 20), (20, 32)))
     np.testing.assert_array_equal(summed, whole)
+
+
+# --- reduce oracle: decomposition laws ------------------------------------
+
+from oracle.reduce import scrunch  # noqa: E402
+
+
+@settings(max_examples=30, deadline=None)
+@given(vals=st.lists(st.integers(-100, 100), min_size=8, max_size=64),
+       factor=st.sampled_from([2, 4]))
+def test_scrunch_sum_composes(vals, factor):
+    # scrunch by a*b == scrunch by a then by b (sum is associative)
+    n = (len(vals) // (factor * 2)) * (factor * 2)
+    if n == 0:
+        return
+    x = np.array(vals[:n], dtype=np.float32)
+    once = scrunch(x, factor * 2, 0, "sum")
+    twice = scrunch(scrunch(x, factor, 0, "sum"), 2, 0, "sum")
+    np.testing.assert_allclose(once, twice, rtol=1e-6)
+
+
+@settings(max_examples=30, deadline=None)
+@given(vals=st.lists(st.integers(-100, 100), min_size=4, max_size=64))
+def test_scrunch_mean_of_whole_axis(vals):
+    n = (len(vals) // 4) * 4
+    if n == 0:
+        return
+    x = np.array(vals[:n], dtype=np.float32)
+    np.testing.assert_allclose(scrunch(x, None, 0, "mean"),
+                               [x.mean()], rtol=1e-5)
+    np.testing.assert_allclose(scrunch(x, None, 0, "max"), [x.max()])
+    np.testing.assert_allclose(scrunch(x, None, 0, "min"), [x.min()])
+
+
+@settings(max_examples=30, deadline=None)
+@given(vals=st.lists(st.integers(-50, 50), min_size=8, max_size=64))
+def test_scrunch_pwrsum_is_sum_of_squared_magnitudes(vals):
+    n = (len(vals) // 4) * 4
+    if n < 8:
+        return
+    re = np.array(vals[:n // 2], dtype=np.float32)
+    im = np.array(vals[n // 2:n], dtype=np.float32)
+    z = (re + 1j * im).astype(np.complex64)
+    got = scrunch(z, 2, 0, "pwrsum")
+    want = (np.abs(z) ** 2).reshape(-1, 2).sum(axis=1)
+    np.testing.assert_allclose(got, want, rtol=1e-5)
