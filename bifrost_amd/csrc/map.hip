@@ -26,6 +26,7 @@
 #include <sys/stat.h>
 #include <cstring>
 #include <list>
+#include <memory>
 #include <mutex>
 #include <sstream>
 #include <string>
@@ -192,24 +193,31 @@ bool dtype_supported(BFdtype dt) {
     return true;
 }
 
+// Entries are reference-counted: a thread launching a kernel holds a
+// shared_ptr copied out under the lock, so eviction (or bfMapClearCache)
+// from another thread cannot hipModuleUnload a module with a launch in
+// flight — the unload happens in the destructor when the last user drops.
 struct CacheEntry {
     hipModule_t module = nullptr;
     hipFunction_t func = nullptr;
+    CacheEntry() = default;
+    CacheEntry(const CacheEntry&) = delete;
+    CacheEntry& operator=(const CacheEntry&) = delete;
+    ~CacheEntry() {
+        if (module) (void)hipModuleUnload(module);
+    }
 };
+using CacheRef = std::shared_ptr<CacheEntry>;
 
 std::mutex g_cache_mutex;
-std::unordered_map<std::string, CacheEntry> g_cache;
+std::unordered_map<std::string, CacheRef> g_cache;
 std::list<std::string> g_cache_order;
 
 void cache_evict_locked() {
     while (g_cache.size() > BF_MAP_KERNEL_CACHE_SIZE && !g_cache_order.empty()) {
         auto key = g_cache_order.front();
         g_cache_order.pop_front();
-        auto it = g_cache.find(key);
-        if (it != g_cache.end()) {
-            hipModuleUnload(it->second.module);
-            g_cache.erase(it);
-        }
+        g_cache.erase(key);  // module unloads when the last ref drops
     }
 }
 
@@ -274,14 +282,15 @@ void disk_cache_store(const std::string& src,
 }
 
 BFstatus load_module(const std::vector<char>& code, const std::string& src,
-                     CacheEntry* out) {
-    CacheEntry e;
-    if (hipModuleLoadData(&e.module, code.data()) != hipSuccess)
+                     CacheRef* out) {
+    auto e = std::make_shared<CacheEntry>();
+    if (hipModuleLoadData(&e->module, code.data()) != hipSuccess) {
+        e->module = nullptr;
         return BF_STATUS_DEVICE_ERROR;
-    if (hipModuleGetFunction(&e.func, e.module, "bfmap_kernel") !=
+    }
+    if (hipModuleGetFunction(&e->func, e->module, "bfmap_kernel") !=
         hipSuccess) {
-        hipModuleUnload(e.module);
-        return BF_STATUS_DEVICE_ERROR;
+        return BF_STATUS_DEVICE_ERROR;  // dtor unloads
     }
     std::lock_guard<std::mutex> lk(g_cache_mutex);
     g_cache[src] = e;
@@ -291,7 +300,7 @@ BFstatus load_module(const std::vector<char>& code, const std::string& src,
     return BF_STATUS_SUCCESS;
 }
 
-BFstatus compile_and_cache(const std::string& src, CacheEntry* out) {
+BFstatus compile_and_cache(const std::string& src, CacheRef* out) {
     {
         std::lock_guard<std::mutex> lk(g_cache_mutex);
         auto it = g_cache.find(src);
@@ -355,13 +364,24 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
     // reference (the elementwise path, constant-folded offsets).
     bool indexed[16] = {false};
     bool any_indexed = false;
+    // A match only counts with a non-identifier character on the LEFT (an
+    // arg named 'a' must not match inside "data(..)"), mirroring the bare
+    // '_' word-boundary check below.
+    auto find_word = [&](const char* hay, const std::string& pat) {
+        for (const char* p = std::strstr(hay, pat.c_str()); p;
+             p = std::strstr(p + 1, pat.c_str())) {
+            if (p == hay ||
+                (!isalnum((unsigned char)p[-1]) && p[-1] != '_'))
+                return true;
+        }
+        return false;
+    };
     for (int a = 0; a < narg; ++a) {
         std::string pat1 = std::string(arg_names[a]) + "(";
         std::string pat2 = std::string(arg_names[a]) + ".shape";
         std::string pat3 = std::string(arg_names[a]) + " (";
-        if (std::strstr(func, pat1.c_str()) ||
-            std::strstr(func, pat2.c_str()) ||
-            std::strstr(func, pat3.c_str())) {
+        if (find_word(func, pat1) || find_word(func, pat2) ||
+            find_word(func, pat3)) {
             indexed[a] = true;
             any_indexed = true;
         }
@@ -515,7 +535,7 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
     os << "    " << func << ";\n";
     os << "  }\n}\n";
 
-    CacheEntry entry;
+    CacheRef entry;
     BF_CHECK(compile_and_cache(os.str(), &entry));
 
     // ---- launch ----
@@ -528,7 +548,7 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
     unsigned blocks = (unsigned)std::min<long>((n + 255) / 256, 32768L);
     if (blocks == 0) blocks = 1;
     hipError_t err = hipModuleLaunchKernel(
-        entry.func, blocks, 1, 1, 256, 1, 1, 0, bfamd::thread_stream(),
+        entry->func, blocks, 1, 1, 256, 1, 1, 0, bfamd::thread_stream(),
         kargs, nullptr);
     BF_CHECK_HIP(err);
     return BF_STATUS_SUCCESS;
@@ -536,8 +556,7 @@ extern "C" BFstatus bfMap(int ndim, long const* shape,
 
 extern "C" BFstatus bfMapClearCache() {
     std::lock_guard<std::mutex> lk(g_cache_mutex);
-    for (auto& kv : g_cache) hipModuleUnload(kv.second.module);
-    g_cache.clear();
+    g_cache.clear();  // each module unloads when its last user drops
     g_cache_order.clear();
     return BF_STATUS_SUCCESS;
 }

@@ -236,6 +236,16 @@ __global__ __launch_bounds__(256) void reduce_cpow_wave_kernel(
 
 // ---- launch plumbing -------------------------------------------------------
 
+// The VEC4 fast path does 16-B dwordx4 loads at in + ioff; a base pointer
+// check alone is not enough — per-output offsets built from sliced/padded
+// outer-dim strides can break the 16-B alignment.  Require every outer
+// stride to be a multiple of 4 elements.
+inline bool vec4_strides_ok(const ReduceArgs& args) {
+    for (int d = 0; d < args.ndim; ++d)
+        if (args.istrides[d] % 4 != 0) return false;
+    return true;
+}
+
 inline int thread_blocks(size_t nout) {
     size_t b = (nout + 255) / 256;
     if (b > 16384) b = 16384;  // grid-strided beyond this (>> 256 WGs)
@@ -256,7 +266,8 @@ void launch_real(const void* in, void* out, const ReduceArgs& args,
                            dim3(wave_blocks(nout)), dim3(256), 0, s,
                            (const IT*)in, (float*)out, args, nout);
     else if (std::is_same<IT, float>::value && args.istride_r == 1 &&
-             args.n % 4 == 0 && ((uintptr_t)in % 16 == 0))
+             args.n % 4 == 0 && ((uintptr_t)in % 16 == 0) &&
+             vec4_strides_ok(args))
         hipLaunchKernelGGL(
             (reduce_real_thread_kernel<IT, OPK, POWER, true>),
             dim3(thread_blocks(nout)), dim3(256), 0, s, (const IT*)in,

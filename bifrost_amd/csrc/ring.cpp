@@ -13,6 +13,8 @@
 #include <bifrost/memory.h>
 #include <bifrost/ring.h>
 
+#include <hip/hip_runtime.h>
+
 #include <algorithm>
 #include <condition_variable>
 #include <cstring>
@@ -23,6 +25,7 @@
 #include <string>
 #include <vector>
 
+#include "hipctx.hpp"
 #include "status.hpp"
 
 namespace {
@@ -92,7 +95,9 @@ struct BFring_impl {
         BFsize pos = abs % capacity;
         if (pos + size <= capacity) return BF_STATUS_SUCCESS;
         BFsize overhang = pos + size - capacity;
-        return bfMemcpy(buf + capacity, space, buf, space, overhang);
+        BFstatus st = bfMemcpy(buf + capacity, space, buf, space, overhang);
+        if (st != BF_STATUS_SUCCESS) return st;
+        return ghost_fence();
     }
     // Propagate bytes written into the ghost area back to the buffer start
     // after a wrapped write span commits.  Caller holds the mutex.
@@ -100,7 +105,21 @@ struct BFring_impl {
         BFsize pos = abs % capacity;
         if (pos + size <= capacity) return BF_STATUS_SUCCESS;
         BFsize overhang = pos + size - capacity;
-        return bfMemcpy(buf, space, buf + capacity, space, overhang);
+        BFstatus st = bfMemcpy(buf, space, buf + capacity, space, overhang);
+        if (st != BF_STATUS_SUCCESS) return st;
+        return ghost_fence();
+    }
+    // Ghost fix-up copies on device-space rings are enqueued async on the
+    // CALLING thread's HIP stream; a peer thread with a different stream
+    // (bfStreamSet / torch-stream interop) could otherwise observe the span
+    // before the copy lands.  Synchronize before publishing.
+    BFstatus ghost_fence() const {
+        if (space == BF_SPACE_CUDA || space == BF_SPACE_CUDA_MANAGED ||
+            space == BF_SPACE_CUDA_HOST) {
+            if (hipStreamSynchronize(bfamd::thread_stream()) != hipSuccess)
+                return BF_STATUS_DEVICE_ERROR;
+        }
+        return BF_STATUS_SUCCESS;
     }
 };
 
@@ -508,7 +527,14 @@ BFstatus bfRingSpanReserve(BFwspan* span, BFring ring,
     // Make sure a wrapped write span reflects current buffer-start contents
     // (partial commits elsewhere could otherwise be clobbered by flush).
     BFstatus st = ring->refresh_ghost(ws->begin, size);
-    if (st != BF_STATUS_SUCCESS) { delete ws; return st; }
+    if (st != BF_STATUS_SUCCESS) {
+        // Roll back the reservation so the window isn't leaked until an
+        // unrelated commit resets reserve_head.
+        ring->reserve_head -= size;
+        delete ws;
+        ring->cv.notify_all();
+        return st;
+    }
     ++ring->nwrite_open;
     *span = ws;
     return BF_STATUS_SUCCESS;
