@@ -884,6 +884,182 @@ void cherk_ci8_mfma_wave_kernel(long n, long k, long nbatch, float alpha,
     }
 }
 
+/* -------- rs2: 4-wave 128x128-complex big-tile register-staged cherk ----- */
+// Round-2 redesign of the rs kernel attacking its measured LDS-read bound
+// (profiles/round1_cherk.md sched-5 ablation: compute loop alone = 2x the
+// MFMA roofline because the 2x2 wave grid reads each operand half twice).
+// Here the workgroup tile is 128x128 complex and each wave owns a 64x64
+// complex quadrant = an 8x8 grid of 16x16x64 i8 MFMAs:
+//   - LDS fragment bytes per MFMA HALVE (perimeter/area: 16 frags feed 64
+//     MFMAs vs 8 frags feeding 16);
+//   - staged bytes per MFMA halve (two 256-B strips feed 256 MFMAs/WG);
+//   - barriers per MFMA drop 4x (one per 64-k slab per 256 WG-MFMAs).
+// The price is 256 accumulator VGPRs per lane (64 x v4i), which the gfx950
+// unified 512-register file carries at 1 wave/SIMD (MI355X_MICROARCH.md
+// "Register files": no spill through 450).  The 64-MFMA burst (~20
+// cyc/SIMD each back-to-back) is long enough to self-hide the tr8 reads
+// pipelined into it; staged global loads prefetch two slabs ahead as in rs.
+// Strips are [64 k][256 B] with a 4-bit chunk XOR swizzle (row stride 256 B
+// = a full bank row; position = chunk ^ (row & 15), the rs8 I-strip
+// pattern).  Requires n%128==0, k%64==0, k>=128, 16-B-aligned a/lda/batch.
+// SCHED: 0 = reads -> writes -> loads -> pipelined burst (early writes
+// overlap the burst; the vmcnt park on stg is ~0 at 2-slab prefetch
+// distance); 1 = pipelined burst first, then writes+loads (rs sched-5
+// order).  Same-box ABAB decides the default.
+template <int SCHED>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(1)))
+void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
+                               const signed char* __restrict__ a, long lda,
+                               long a_b, float beta, f2* __restrict__ c,
+                               long c_row, long c_b, long ntiles) {
+    __shared__ signed char lds[2][2][64][256];  // [buf][strip][k][byte]
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;
+
+    int tr_row = 8 * (lane >> 4) + ((lane & 15) >> 1);
+    int tr_half = lane & 1;
+
+    // staging: 128 threads per strip; thread covers 128 contiguous global
+    // bytes of one k-row (row tt>>1, byte half 128*(tt&1)).
+    int st_strip = tid >> 7;
+    int tt = tid & 127;
+    int st_row = tt >> 1;
+    int st_cq = (tt & 1) * 8;  // first of 8 consecutive 16-B chunks
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        long bi, bj;
+        lift_tri(t, bi, bj);
+        long i0 = bi * 128, j0 = bj * 128;
+        bool diag = bi == bj;
+        // Diagonal tiles: the (wr=0,wc=1) quadrant lies entirely above the
+        // diagonal (skips MFMAs, still stages); in wr==wc quadrants the
+        // ta<tb MFMA tiles are above the diagonal and skip too.
+        bool skip_all = diag && wr < wc;
+        bool diag_q = diag && wr == wc;
+        v4i acc[8][8];
+#pragma unroll
+        for (int x = 0; x < 8; ++x)
+#pragma unroll
+            for (int y = 0; y < 8; ++y) acc[x][y] = v4i{};
+
+        long base_col = st_strip ? j0 : i0;
+        const long slab_step = 64 * lda * 2;
+        const signed char* src0 = ab + (long)st_row * lda * 2 +
+                                  base_col * 2 + 128 * (tt & 1);
+        v4i stg[8];
+        const signed char* load_next = src0;
+        auto load_slab = [&]() {
+            const v4i* pv = (const v4i*)__builtin_assume_aligned(
+                load_next, 16);
+            load_next += slab_step;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) stg[e] = pv[e];
+        };
+        auto write_slab = [&](int buf) {
+            int swz = st_row & 15;
+            signed char* base = &lds[buf][st_strip][st_row][0];
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                *(v4i*)(base + 16 * ((st_cq + e) ^ swz)) = stg[e];
+        };
+        auto frag = [&](const signed char* base, int cc) {
+            const signed char* p = base + tr_row * 256 +
+                16 * (cc ^ (tr_row & 15)) + 8 * tr_half;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 32 * 256));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        // 8x8 MFMA burst over the wave quadrant, fa two-rows-ahead
+        // pipelined (the rs sched-5 pattern at twice the width).
+        auto burst = [&](int buf) {
+            if (skip_all) return;
+            const signed char* bI = &lds[buf][0][0][0];
+            const signed char* bJ = &lds[buf][1][0][0];
+            v4i fb[8];
+#pragma unroll
+            for (int tb = 0; tb < 8; ++tb)
+                fb[tb] = frag(bJ, 8 * wc + tb);
+            v4i fa0 = frag(bI, 8 * wr);
+            v4i fa1 = frag(bI, 8 * wr + 1);
+#pragma unroll
+            for (int ta = 0; ta < 8; ++ta) {
+                v4i fa2;
+                if (ta < 6) fa2 = frag(bI, 8 * wr + ta + 2);
+#pragma unroll
+                for (int tb = 0; tb < 8; ++tb) {
+                    if (diag_q && ta < tb) continue;
+                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                        fa0, fb[tb], acc[ta][tb], 0, 0, 0);
+                }
+                fa0 = fa1;
+                fa1 = fa2;
+            }
+        };
+
+        int nslab = (int)(k / 64);
+        load_slab();
+        write_slab(0);
+        if (nslab > 1) load_slab();
+        __syncthreads();
+        for (int s = 0; s < nslab; ++s) {
+            int buf = s & 1;
+            if (SCHED == 0) {
+                // early writes: they wait only the 2-slab-old stg loads
+                // (long landed) and drain across the MFMA burst.
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+                burst(buf);
+            } else {
+                burst(buf);
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+            }
+            __syncthreads();
+        }
+#pragma unroll
+        for (int ta = 0; ta < 8; ++ta) {
+#pragma unroll
+            for (int tb = 0; tb < 8; ++tb) {
+                if (skip_all || (diag_q && ta < tb)) continue;
+                long arow0 = i0 + 64 * wr + 8 * ta;
+                long acol = j0 + 64 * wc + 8 * tb + ((lane & 15) >> 1);
+#pragma unroll
+                for (int p = 0; p < 2; ++p) {
+                    int v0 = acc[ta][tb][2 * p];
+                    int v1 = acc[ta][tb][2 * p + 1];
+                    int sv0 = __shfl_xor(v0, 1);
+                    int sv1 = __shfl_xor(v1, 1);
+                    long i = arow0 + 2 * (lane >> 4) + p;
+                    long j = acol;
+                    bool write = (lane & 1) == 0 && i < n && j < n && i >= j;
+                    if (write) {
+                        float re = (float)(v0 + sv1);
+                        float im = (float)(sv0 - v1);
+                        f2 prev = beta != 0.f ? cb[i * c_row + j] : f2{};
+                        cb[i * c_row + j] = f2{alpha * re + beta * prev.x,
+                                               alpha * im + beta * prev.y};
+                    }
+                }
+            }
+        }
+    }
+}
+
 /* -------- 8-wave rectangular-tile register-staged cherk (rs8) ----------- */
 // 512 threads / 8 waves per 128x64-complex output tile (wave grid 4x2,
 // each wave the same 32x32-complex quadrant as the rs kernel).  Staged
@@ -1627,10 +1803,35 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                         ((a_k * 2) % 16 == 0) && ((a_b * 2) % 16 == 0);
             bool want_rs = !sel || strcmp(sel, "rs") == 0;
             bool want_wave = sel && strcmp(sel, "wave") == 0;
+            // rs2 big-tile kernel: default when eligible (round-2 winner);
+            // BIFROST_CHERK=rs forces the round-1 kernel.
+            bool want_rs2 = (!sel || strcmp(sel, "rs2") == 0) &&
+                            n % 128 == 0;
             // rs8 (8-wave 128x64 tile) measures ~equal to rs (1.23 vs
             // 1.25 Gsamp/s at config 3) — opt-in until it wins.
             const char* sel8 = getenv("BIFROST_CHERK");
             bool want_rs8 = sel8 && strcmp(sel8, "rs8") == 0;
+            if (al16 && want_rs2) {
+                long nb2 = n / 128;
+                long ntiles2 = nb2 * (nb2 + 1) / 2;
+                long nflat2 = ((ntiles2 * nbatch + 7) / 8) * 8;
+                dim3 grid2(cap_grid(nflat2, 65535), 1);
+                const char* schenv = getenv("BIFROST_CHERK_SCHED");
+                int sched = schenv ? atoi(schenv) : 0;
+                auto launch_rs2 = [&](auto kern) {
+                    hipLaunchKernelGGL(kern, grid2, dim3(256), 0, stream, n,
+                                       k, nbatch, (float)alpha,
+                                       (const signed char*)a, a_k, a_b,
+                                       (float)beta, (f2*)c, c_row, c_b,
+                                       ntiles2);
+                };
+                if (sched == 1)
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<1>);
+                else
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<0>);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
             if (al16 && want_rs8 && n % 128 == 0) {
                 long nti = n / 64;           // 64-wide col blocks
                 long nI = n / 128;
