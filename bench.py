@@ -76,31 +76,86 @@ def flops_per_step(ntime, nchan, n):
     return ntime * nchan * cmacs * 8
 
 
-def cpu_baseline(sample_nchan=4, sample_ntime=1024):
-    """The oracle (numpy/BLAS restatement) timed on this host's cores.
-    Bounded sample, scaled to samples/sec."""
-    import oracle
+def physical_cores():
+    """Physical (not SMT-logical) core count, as BASELINE.md promised."""
+    try:
+        import psutil
+        c = psutil.cpu_count(logical=False)
+        if c:
+            return c
+    except Exception:
+        pass
+    return os.cpu_count()
+
+
+def cpu_baseline(ntime=NTIME, max_seconds=12.0):
+    """The oracle (numpy/BLAS restatement of the reference CPU linalg
+    path) timed on this host's cores at the FULL config-3 per-channel
+    shape: per channel, X^H @ X with X (ntime=4096, n=512) complex64 —
+    a single large 2-D cgemm per channel, which provably engages BLAS
+    threading (unlike the batched 3-D matmul the round-1 leg used).
+    Loops channels until ~max_seconds of CPU work, then scales."""
     from oracle.linalg import H
 
-    x8 = make_voltages(sample_ntime, sample_nchan, N, seed=99)
-    x = x8.astype(np.float32).view(np.complex64).reshape(
-        sample_ntime, sample_nchan, N).transpose(1, 0, 2)
-    x = np.ascontiguousarray(x)
-    # warm + time
-    _ = np.matmul(H(x[:1]), x[:1])
+    x8 = make_voltages(ntime, 1, N, seed=99)
+    xc = np.ascontiguousarray(
+        x8.astype(np.float32).view(np.complex64).reshape(ntime, N))
+    xh = np.ascontiguousarray(H(xc))
+    # warm BLAS thread pool on a slice
+    _ = xh[:, :256] @ xc[:256]
     t0 = time.perf_counter()
-    c = np.matmul(H(x), x)
+    done = 0
+    while done < NCHAN_PER_GPU:
+        c = xh @ xc
+        done += 1
+        if time.perf_counter() - t0 > max_seconds:
+            break
     dt = time.perf_counter() - t0
-    assert c.shape == (sample_nchan, N, N)
-    samples = sample_ntime * sample_nchan
+    assert c.shape == (N, N)
+    samples = done * ntime
     return {
         "value": samples / dt / 1e9,
         "unit": "Gsamp/s",
-        "cores": os.cpu_count(),
+        "cores": physical_cores(),
         "kind": "port",
-        "sample": "numpy/BLAS oracle, %d chan x %d time, n=%d"
-                  % (sample_nchan, sample_ntime, N),
+        "sample": "numpy/BLAS oracle, %d x full config-3 per-channel "
+                  "cgemm (ntime=%d, n=%d) in %.1fs"
+                  % (done, ntime, N, dt),
     }
+
+
+def traffic_bytes_per_launch(workload_key):
+    """Measured HBM/fabric FETCH bytes per cherk launch, from separate
+    rocprofv3 --pmc passes (TCC_EA0_RDREQ_sum x 64 x 2 per the gfx950
+    FETCH_SIZE calibration, MI355X_MICROARCH.md §HBM; summaries under
+    profiles/).  The committed value lives in profiles/
+    traffic_manifest.json KEYED BY the sha of the kernel source and the
+    kernel-selection env, so it self-invalidates the first time the
+    kernel changes without a re-profile (returns None -> traffic: null).
+    BIFROST_TRAFFIC_BYTES_PER_LAUNCH overrides after a fresh profile."""
+    env = os.environ.get("BIFROST_TRAFFIC_BYTES_PER_LAUNCH")
+    if env:
+        return float(env)
+    here = os.path.dirname(os.path.abspath(__file__))
+    src = os.path.join(here, "bifrost_amd", "csrc", "linalg.hip")
+    man = os.path.join(here, "profiles", "traffic_manifest.json")
+    try:
+        import hashlib
+        with open(src, "rb") as f:
+            sha16 = hashlib.sha256(f.read()).hexdigest()[:16]
+        with open(man) as f:
+            entries = json.load(f)["entries"]
+    except (OSError, KeyError, ValueError):
+        return None
+    sel = os.environ.get("BIFROST_CHERK") or None
+    sched = os.environ.get("BIFROST_CHERK_SCHED") or None
+    for e in entries:
+        if (e.get("workload") == workload_key and
+                e.get("kernel_src_sha16") == sha16 and
+                e.get("cherk_env") == sel and
+                e.get("sched_env") == sched):
+            return float(e["fetch_bytes_per_launch"])
+    return None
 
 
 def main():
@@ -203,15 +258,8 @@ def main():
     per_launch_s = (kernel_ms / 1e3) / args.steps
     alg_bytes = algorithmic_bytes_per_step(ntime, nchan, N)
     achieved_gbs = alg_bytes / per_launch_s / 1e9
-    # Measured HBM/fabric traffic per launch comes from separate rocprofv3
-    # --pmc passes (TCC_EA0_RDREQ_sum x 64 x 2 per the gfx950 FETCH_SIZE
-    # calibration; see profiles/).  The default is the committed
-    # measurement for the shipped sched-5 kernel at the default workload
-    # (profiles/round1_cherk.md: sched-5 FETCH 5.15 GB/launch, L2 hit 73%);
-    # override with BIFROST_TRAFFIC_BYTES_PER_LAUNCH after re-profiling.
-    traffic_env = os.environ.get("BIFROST_TRAFFIC_BYTES_PER_LAUNCH")
-    if traffic_env is None and ntime == 4096 and nchan == 512 and N == 512:
-        traffic_env = "5.15e9"
+    workload_key = "xcorr_n%d_c%d_t%d" % (N, nchan, ntime)
+    traffic = traffic_bytes_per_launch(workload_key)
     # The binding resource for the cherk kernel at n=512 is the i8 MFMA
     # pipe, not HBM: algorithmic intensity = 2n real-OPS per input byte =
     # 1024 OPS/B, above the machine balance (~7.9e15 int-OPS/s over 8e12
@@ -226,7 +274,7 @@ def main():
         "peak": MFMA_I8_PEAK_TOPS,
         "unit": "TFLOP/s",
         "frac": round(achieved_tops / MFMA_I8_PEAK_TOPS, 4),
-        "traffic": float(traffic_env) if traffic_env else None,
+        "traffic": traffic,
         "hbm_achieved_gbs": round(achieved_gbs, 1),
         "hbm_frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
     }

@@ -86,6 +86,80 @@ class TestCorrelatorKernel:
         # depends on it)
         run_corr(linalg, 64, 16, 3, beta=1.0)
 
+    def test_beta_widened(self, linalg):
+        # fractional, >1 and negative beta on both MFMA kernels: n=64
+        # (nstand 32 -> rs kernel) and n=128 (nstand 64 -> rs2 kernel)
+        for beta in (0.5, 2.5, -1.0):
+            run_corr(linalg, 128, 32, 2, beta=beta)
+            run_corr(linalg, 128, 64, 2, beta=beta)
+        # and on the odd-shape (non-MFMA) correlator path
+        run_corr(linalg, 65, 17, 2, beta=0.5)
+
+    def test_ci8_minus_128(self, linalg):
+        # The reference's own test excludes -128 (test_linalg.py:55); our
+        # kernels accumulate exactly in i32, so -128 is handled with no
+        # special-casing.  Pin that behavior (both MFMA kernels + edges).
+        np.random.seed(7)
+        for ntime, nstand, nchan in [(64, 32, 2), (64, 64, 2), (33, 17, 1)]:
+            x_shape = (ntime, nchan, nstand * 2)
+            x8 = np.random.randint(-128, 128, size=x_shape + (2,)) \
+                .astype(np.int8)
+            # force some -128s in deterministically
+            x8[0, 0, 0, 0] = -128
+            x8[-1, -1, -1, 1] = -128
+            x = x8.astype(np.float32).view(np.complex64).reshape(x_shape)
+            xv = x.transpose(1, 0, 2)
+            b_gold = np.matmul(H(xv), xv)
+            triu = np.triu_indices(xv.shape[-1], 1)
+            b_gold[..., triu[0], triu[1]] = 0
+            xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                       .reshape(x_shape)), space="cuda")
+            b = bf.zeros_like(b_gold, space="cuda")
+            linalg.matmul(1, None, xb.transpose(1, 0, 2), 0, b)
+            np.testing.assert_allclose(np.asarray(b.copy("system")),
+                                       b_gold, RTOL * 10, ATOL)
+
+
+class TestGenericBeta:
+    """beta on the generic (non-specialized) herk/gemm dtype paths —
+    round-2 widening of the one-point beta coverage."""
+
+    def _herk_cf32(self, linalg, beta):
+        np.random.seed(21)
+        nchan, ntime, n = 2, 48, 24
+        x = (np.random.standard_normal((nchan, ntime, n, 2))
+             .astype(np.float32)).view(np.complex64)[..., 0]
+        gold = np.matmul(H(x), x)
+        triu = np.triu_indices(n, 1)
+        gold[..., triu[0], triu[1]] = 0
+        xb = bf.asarray(x, space="cuda")
+        b = bf.zeros_like(gold, space="cuda")
+        linalg.matmul(1, None, xb, 0, b)
+        linalg.matmul(1, None, xb, beta, b)   # c = X + beta*X
+        np.testing.assert_allclose(np.asarray(b.copy("system")),
+                                   gold * (1 + beta), RTOL, ATOL)
+
+    def _gemm_cf32(self, linalg, beta):
+        np.random.seed(22)
+        m, k, n = 17, 31, 23
+        a = (np.random.standard_normal((m, k, 2)).astype(np.float32)
+             .view(np.complex64))[..., 0]
+        bmat = (np.random.standard_normal((k, n, 2)).astype(np.float32)
+                .view(np.complex64))[..., 0]
+        gold = a @ bmat
+        ab = bf.asarray(a, space="cuda")
+        bb = bf.asarray(bmat, space="cuda")
+        c = bf.zeros_like(gold, space="cuda")
+        linalg.matmul(1, ab, bb, 0, c)
+        linalg.matmul(1, ab, bb, beta, c)
+        np.testing.assert_allclose(np.asarray(c.copy("system")),
+                                   gold * (1 + beta), RTOL, ATOL * 10)
+
+    def test_generic_beta(self, linalg):
+        for beta in (0.5, 2.0, -0.75):
+            self._herk_cf32(linalg, beta)
+            self._gemm_cf32(linalg, beta)
+
 
 class TestBeamformerKernel:
     def test_small_sweep(self, linalg):

@@ -1,0 +1,117 @@
+#!/usr/bin/env python3
+"""Same-box ABAB cherk kernel A/B (round-1 methodology rule: box-to-box
+sustained clocks vary up to ~25%, so kernel comparisons are only valid
+env-switched and interleaved within ONE process on ONE box).
+
+Usage: python tools/ab_cherk.py [variant ...]
+  variant = NAME:CHERK_ENV:SCHED_ENV, e.g. rs2s0:rs2:0  rs:rs:5
+  default sweep: rs2 sched 0/1 vs the round-1 rs default.
+
+Runs the full config-3 workload (n=512, nchan=512, ntime=4096) with a
+fresh parity check per variant against a small-slice oracle, REPS
+interleaved reps of STEPS timed steps each, and prints one JSON line per
+(variant, rep) plus a summary.
+"""
+
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+REPS = 3
+STEPS = 10
+WARM = 3
+
+
+def main():
+    variants = []
+    for spec in sys.argv[1:]:
+        name, cherk, sched = spec.split(":")
+        variants.append((name, cherk or None, sched or None))
+    if not variants:
+        variants = [
+            ("rs2s0", "rs2", "0"),
+            ("rs2s1", "rs2", "1"),
+            ("rs", "rs", "5"),
+        ]
+
+    import torch
+    import bifrost_amd as bf
+    from bifrost_amd import device as bf_device
+    from bifrost_amd.linalg import LinAlg
+    from oracle.linalg import H
+
+    bf_device.set_device(0)
+    torch.cuda.set_device(0)
+    bf_device.set_stream(torch.cuda.current_stream().cuda_stream)
+
+    N, NCHAN, NTIME = 512, 512, 4096
+    rng = np.random.RandomState(1234)
+    x8 = rng.randint(-127, 128, size=(NTIME, NCHAN, N, 2), dtype=np.int8)
+    x_dev = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                  .reshape(NTIME, NCHAN, N)), space="cuda")
+    x_view = x_dev.transpose(1, 0, 2)
+    vis = bf.ndarray(space="cuda", shape=(NCHAN, N, N), dtype="cf32")
+    linalg = LinAlg()
+
+    # small-slice parity gold (first 2 channels, full k)
+    xs = x8[:, :2].astype(np.float32).view(np.complex64) \
+        .reshape(NTIME, 2, N).transpose(1, 0, 2)
+    gold2 = np.matmul(H(xs), xs)
+    tri = np.triu_indices(N, 1)
+    gold2[..., tri[0], tri[1]] = 0
+
+    def set_env(cherk, sched):
+        for k in ("BIFROST_CHERK", "BIFROST_CHERK_SCHED"):
+            os.environ.pop(k, None)
+        if cherk:
+            os.environ["BIFROST_CHERK"] = cherk
+        if sched:
+            os.environ["BIFROST_CHERK_SCHED"] = sched
+
+    results = {}
+    for name, cherk, sched in variants:
+        set_env(cherk, sched)
+        linalg.matmul(1, None, x_view, 0, vis)
+        torch.cuda.synchronize()
+        got = np.asarray(vis.copy("system"))[:2]
+        ok = np.allclose(got.view(np.float32), gold2.view(np.float32),
+                         rtol=1e-3, atol=1e-2)
+        results[name] = {"parity": bool(ok), "gsps": []}
+        if not ok:
+            bad = np.abs(got.view(np.float32) - gold2.view(np.float32))
+            print(json.dumps({"variant": name, "parity": False,
+                              "max_abs_err": float(bad.max())}))
+
+    for rep in range(REPS):
+        for name, cherk, sched in variants:
+            if not results[name]["parity"]:
+                continue
+            set_env(cherk, sched)
+            for _ in range(WARM):
+                linalg.matmul(1, None, x_view, 1, vis)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(STEPS):
+                linalg.matmul(1, None, x_view, 1, vis)
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            gsps = STEPS * NTIME * NCHAN / dt / 1e9
+            results[name]["gsps"].append(round(gsps, 4))
+            print(json.dumps({"variant": name, "rep": rep,
+                              "ms_per_step": round(dt / STEPS * 1e3, 4),
+                              "gsamp_per_s": round(gsps, 4)}))
+
+    summary = {n: {"parity": r["parity"],
+                   "gsps_min": min(r["gsps"]) if r["gsps"] else None,
+                   "gsps_max": max(r["gsps"]) if r["gsps"] else None}
+               for n, r in results.items()}
+    print("SUMMARY " + json.dumps(summary))
+
+
+if __name__ == "__main__":
+    main()
