@@ -108,6 +108,60 @@ template<typename T>
 __device__ Complex<T> conj(const Complex<T>& c) {
     return Complex<T>(c.real, -c.imag);
 }
+// Packed 4+4-bit complex: ONE byte per element (re in the HIGH nibble,
+// the ci4 convention of src/Complex.hpp:149-168), so it is byte-addressed
+// like every other dtype.  Participates in map expressions as a value
+// convertible to/from Complex<T>; direct .real/.imag member access is not
+// available on the packed form (use cf32_t(a).real etc.).
+struct ci4_t {
+    signed char b;
+    __device__ signed char real_get() const { return (signed char)(b >> 4); }
+    __device__ signed char imag_get() const {
+        return (signed char)((signed char)(b << 4) >> 4);
+    }
+    template<typename U>
+    __device__ operator Complex<U>() const {
+        return Complex<U>((U)real_get(), (U)imag_get());
+    }
+    template<typename U>
+    __device__ ci4_t& operator=(const Complex<U>& o) {
+        int re = (int)o.real, im = (int)o.imag;
+        b = (signed char)(((re & 0xF) << 4) | (im & 0xF));
+        return *this;
+    }
+    __device__ ci4_t& assign(int r, int i) {
+        b = (signed char)(((r & 0xF) << 4) | (i & 0xF));
+        return *this;
+    }
+    __device__ Complex<signed char> conj() const {
+        return Complex<signed char>(real_get(), (signed char)-imag_get());
+    }
+    __device__ int mag2() const {
+        int r = real_get(), i = imag_get();
+        return r * r + i * i;
+    }
+};
+#define CI4_BINOP(op)                                                    \
+template<typename U>                                                     \
+__device__ Complex<U> operator op(const ci4_t& a, const Complex<U>& b) { \
+    return Complex<U>((U)a.real_get(), (U)a.imag_get()) op b;            \
+}                                                                        \
+template<typename U>                                                     \
+__device__ Complex<U> operator op(const Complex<U>& a, const ci4_t& b) { \
+    return a op Complex<U>((U)b.real_get(), (U)b.imag_get());            \
+}
+CI4_BINOP(+)
+CI4_BINOP(-)
+CI4_BINOP(*)
+__device__ inline Complex<float> operator*(const ci4_t& a, float s) {
+    return Complex<float>(a.real_get() * s, a.imag_get() * s);
+}
+__device__ inline Complex<float> operator*(float s, const ci4_t& a) {
+    return a * s;
+}
+__device__ inline Complex<signed char> conj(const ci4_t& a) {
+    return a.conj();
+}
 typedef Complex<signed char> ci8_t;
 typedef Complex<short>       ci16_t;
 typedef Complex<int>         ci32_t;
@@ -184,7 +238,14 @@ std::string dtype_ctype(BFdtype dt) {
 
 bool dtype_supported(BFdtype dt) {
     int nbit = dt & BF_DTYPE_NBIT_BITS;
-    if (nbit < 8) return false;  // packed sub-byte not supported in map
+    bool cplx = dt & BF_DTYPE_COMPLEX_BIT;
+    // ci4 packs 4+4 bits in ONE byte per element — byte-addressable, so
+    // map supports it (as the reference does via Complex<FourBit>).
+    // Plain i4/u4 and narrower (2+ elements per byte) stay unsupported.
+    if (nbit == 4 && cplx) {
+        return (dt & BF_DTYPE_TYPE_BITS) == BF_DTYPE_INT_TYPE;
+    }
+    if (nbit < 8) return false;  // true sub-byte not supported in map
     int type = dt & BF_DTYPE_TYPE_BITS;
     if (type != BF_DTYPE_INT_TYPE && type != BF_DTYPE_UINT_TYPE &&
         type != BF_DTYPE_FLOAT_TYPE)

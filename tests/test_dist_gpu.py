@@ -1,12 +1,21 @@
-"""RCCL on hardware: 2 ranks co-resident on ONE MI355X (round-2 item:
-the only available hardware evidence for the multi-GPU path until the
-driver fields an 8-GPU node).  Exercises torch.distributed init over
-nccl(=RCCL), the channel-shard assignment, and the --time-split
-visibility all-reduce end to end on device: each rank correlates its own
-time half of the SAME channels through the HIP cherk kernel, all-reduces
-the per-channel visibility matrices over RCCL, and the combined result
-must match the full-integration numpy oracle (SURVEY.md §8e;
-bench.py --time-split semantics)."""
+"""RCCL on hardware (round-2): the only multi-GPU evidence available on
+a 1-GPU lease.
+
+Finding (recorded 2026-09, MI355X, RCCL 2.26.6): RCCL REJECTS two ranks
+co-resident on one GPU — `ncclInvalidUsage: Duplicate GPU detected :
+rank 0 and rank 1 both on CUDA device a000` — so the bench's N>1 path
+can only execute with real multi-device nodes (the driver's scaling
+run).  What CAN run here and does:
+  * a world_size=1 nccl(=RCCL) process group driving the EXACT
+    --time-split code path of bench.py on device: init, HIP cherk
+    correlation, dist.all_reduce on the visibility tensor (RCCL launches
+    its reduction kernel even at world 1), barrier, destroy — pinning
+    RCCL init + kernel launch + our stream interop on hardware;
+  * the 2-rank attempt, kept as a probe: it must either pass (future
+    RCCL permitting co-residence) or fail with the known Duplicate-GPU
+    signature, which the test records via skip.
+Combine-semantics at world>1 are covered by the gloo world-2 CPU tests
+(tests/test_dist_cpu.py), which share this code path."""
 
 import os
 import subprocess
@@ -30,7 +39,7 @@ sys.path.insert(0, os.environ["BIFROST_REPO"])
 rank = int(os.environ["RANK"])
 world = int(os.environ["WORLD_SIZE"])
 dist.init_process_group("nccl", rank=rank, world_size=world)
-torch.cuda.set_device(0)  # both ranks co-resident on the one GPU
+torch.cuda.set_device(0)
 
 import bifrost_amd as bf
 from bifrost_amd import device as bf_device
@@ -73,23 +82,21 @@ print("RANK%d_OK" % rank)
 """
 
 
-def test_rccl_two_ranks_one_gpu(tmp_path):
-    script = tmp_path / "rccl_worker.py"
-    script.write_text(_WORKER)
+def _launch(world, port):
     procs = []
-    for rank in range(2):
+    for rank in range(world):
         env = dict(os.environ)
         env.update({
             "MASTER_ADDR": "127.0.0.1",
-            "MASTER_PORT": "29572",
+            "MASTER_PORT": str(port),
             "RANK": str(rank),
-            "WORLD_SIZE": "2",
+            "WORLD_SIZE": str(world),
             "BIFROST_REPO": _REPO,
             # dmabuf IPC (see environment contract); required for RCCL
             "HSA_ENABLE_IPC_MODE_LEGACY": "0",
         })
         procs.append(subprocess.Popen(
-            [sys.executable, str(script)], env=env,
+            [sys.executable, "-c", _WORKER], env=env,
             stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
     outs = []
     for p in procs:
@@ -98,8 +105,30 @@ def test_rccl_two_ranks_one_gpu(tmp_path):
         except subprocess.TimeoutExpired:
             for q in procs:
                 q.kill()
-            pytest.fail("RCCL 2-rank worker timed out")
+            pytest.fail("RCCL worker timed out")
         outs.append(out.decode(errors="replace"))
-    for rank, (p, out) in enumerate(zip(procs, outs)):
-        assert p.returncode == 0, "rank %d failed:\n%s" % (rank, out)
-        assert "RANK%d_OK" % rank in out, out
+    return procs, outs
+
+
+def test_rccl_world1_time_split_path():
+    """RCCL init + all_reduce + cherk on device, world_size=1 (always
+    must pass on a 1-GPU box)."""
+    procs, outs = _launch(1, 29573)
+    assert procs[0].returncode == 0, "rank 0 failed:\n%s" % outs[0]
+    assert "RANK0_OK" in outs[0], outs[0]
+
+
+def test_rccl_two_ranks_one_gpu_probe():
+    """2 co-resident ranks: passes if RCCL permits, else records the
+    known Duplicate-GPU rejection (hardware evidence either way)."""
+    procs, outs = _launch(2, 29574)
+    if all(p.returncode == 0 for p in procs):
+        for rank, out in enumerate(outs):
+            assert "RANK%d_OK" % rank in out, out
+        return
+    joined = "\n".join(outs)
+    if "Duplicate GPU detected" in joined:
+        pytest.skip("RCCL forbids co-resident ranks on one GPU "
+                    "(Duplicate GPU detected) — N>1 needs a multi-GPU "
+                    "node; see docstring")
+    pytest.fail("unexpected failure:\n%s" % joined)

@@ -243,3 +243,43 @@ class TestCi4Astype:
         back = np.asarray(a.astype("ci4").astype("cf32").copy("system"))
         np.testing.assert_array_equal(back, np.array([2 + 1j, -4 - 3j],
                                                      dtype=np.complex64))
+
+
+class TestMapCI4:
+    """Direct bfMap over packed ci4 arrays (round-2: closes the one map
+    dtype hole; ci4 is byte-per-element so map addresses it directly,
+    re in the HIGH nibble per the quantize/linalg convention)."""
+
+    def test_map_read_ci4(self):
+        # unpack-and-scale through map: b = a * s
+        raw = np.array([0x10, 0x32, 0xDC, 0x7F], dtype=np.uint8)
+        a = bf.asarray(bf.ndarray(raw.reshape(4, 1).view(bf.DataType.ci4)),
+                       space="cuda")
+        b = bf.zeros((4, 1), dtype="cf32", space="cuda")
+        bf.map("b = a * s", {"b": b, "a": a, "s": np.float32(2.0)})
+        got = np.asarray(b.copy("system"))[:, 0]
+        # high-nibble re: 0x10 -> 1+0j, 0x32 -> 3+2j, 0xDC -> -3-4j,
+        # 0x7F -> 7-1j
+        np.testing.assert_array_equal(
+            got, 2.0 * np.array([1, 3 + 2j, -3 - 4j, 7 - 1j],
+                                dtype=np.complex64))
+
+    def test_map_write_ci4(self):
+        # quantize through map: b(ci4) = a(ci8)
+        vals = np.zeros((3,), dtype=bf.DataType.ci8)
+        vals["re_im"]["re"] = [1, -3, 7]
+        vals["re_im"]["im"] = [2, -4, -1]
+        a = bf.asarray(bf.ndarray(vals), space="cuda")
+        b = bf.ndarray(shape=(3,), dtype="ci4", space="cuda")
+        bf.map("b = a", {"b": b, "a": a})
+        got = np.asarray(b.copy("system"))["re_im"]
+        np.testing.assert_array_equal(got, [0x12, 0xDC, 0x7F])
+
+    def test_map_ci4_conj_mag2(self):
+        raw = np.array([0x32, 0xDC], dtype=np.uint8)
+        a = bf.asarray(bf.ndarray(raw.reshape(2).view(bf.DataType.ci4)),
+                       space="cuda")
+        m = bf.zeros((2,), dtype="f32", space="cuda")
+        bf.map("m = a.mag2()", {"m": m, "a": a})
+        np.testing.assert_array_equal(np.asarray(m.copy("system")),
+                                      [13.0, 25.0])
