@@ -888,24 +888,35 @@ void cherk_ci8_mfma_wave_kernel(long n, long k, long nbatch, float alpha,
 // Round-2 redesign of the rs kernel attacking its measured LDS-read bound
 // (profiles/round1_cherk.md sched-5 ablation: compute loop alone = 2x the
 // MFMA roofline because the 2x2 wave grid reads each operand half twice).
-// Here the workgroup tile is 128x128 complex and each wave owns a 64x64
+// The workgroup tile is 128x128 complex and each wave owns a 64x64
 // complex quadrant = an 8x8 grid of 16x16x64 i8 MFMAs:
 //   - LDS fragment bytes per MFMA HALVE (perimeter/area: 16 frags feed 64
 //     MFMAs vs 8 frags feeding 16);
-//   - staged bytes per MFMA halve (two 256-B strips feed 256 MFMAs/WG);
+//   - staged bytes per MFMA halve (two strips feed 256 MFMAs/WG);
 //   - barriers per MFMA drop 4x (one per 64-k slab per 256 WG-MFMAs).
-// The price is 256 accumulator VGPRs per lane (64 x v4i), which the gfx950
-// unified 512-register file carries at 1 wave/SIMD (MI355X_MICROARCH.md
-// "Register files": no spill through 450).  The 64-MFMA burst (~20
-// cyc/SIMD each back-to-back) is long enough to self-hide the tr8 reads
-// pipelined into it; staged global loads prefetch two slabs ahead as in rs.
-// Strips are [64 k][256 B] with a 4-bit chunk XOR swizzle (row stride 256 B
-// = a full bank row; position = chunk ^ (row & 15), the rs8 I-strip
-// pattern).  Requires n%128==0, k%64==0, k>=128, 16-B-aligned a/lda/batch.
-// SCHED: 0 = reads -> writes -> loads -> pipelined burst (early writes
-// overlap the burst; the vmcnt park on stg is ~0 at 2-slab prefetch
-// distance); 1 = pipelined burst first, then writes+loads (rs sched-5
-// order).  Same-box ABAB decides the default.
+// The 256 accumulator regs per lane (64 x v4i) land in the AGPR half of
+// the gfx950 unified 512-register file at 1 wave/SIMD.
+//
+// v2 (this form): the first rs2 used XOR chunk swizzles and measured
+// ISSUE-bound at occupancy 1 (profiles/round2_cherk.md: 38% ACTIVE_INST;
+// ~490 non-MFMA instructions per slab, mostly per-access swizzle VALU the
+// register cap left no room to hoist).  Strips are now padded to 272-B
+// rows, which makes the tr8 gather bank-CONFLICT-FREE with NO swizzle
+// (bank = (68r + 4c + 2h) mod 64: distinct for all 32 lanes of each
+// group), so every LDS read/write folds to one per-lane base register
+// plus an immediate offset - near-zero addressing VALU per slab.  Write
+// side interleaves the two 128-B thread-halves chunk-wise (position
+// 2e+half) so the 8 ds_write_b128 are conflict-free too, and the paired
+// global loads coalesce to 32-B row runs.  All 16 fragments load into
+// registers before the 64-MFMA burst (zero mid-burst lgkm waits);
+// diagonal quadrants take a separate reduced burst so interior tiles
+// carry no branches.  Requires n%128==0, k%64==0, k>=128, 16-B-aligned.
+// SCHED: 0 = writes -> loads -> burst (early writes overlap the burst;
+// the vmcnt park on stg is ~0 at 2-slab prefetch distance); 1 = burst
+// first, then writes+loads (rs sched-5 order).
+#define RS2_ROW 272
+#define RS2_STRIP (64 * RS2_ROW)
+#define RS2_BUF (2 * RS2_STRIP)
 template <int SCHED>
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(1)))
@@ -913,7 +924,7 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
                                const signed char* __restrict__ a, long lda,
                                long a_b, float beta, f2* __restrict__ c,
                                long c_row, long c_b, long ntiles) {
-    __shared__ signed char lds[2][2][64][256];  // [buf][strip][k][byte]
+    __shared__ signed char lds[2][2][64][RS2_ROW];  // [buf][strip][k][byte]
     int tid = threadIdx.x;
     int lane = tid & 63;
     int wave = tid >> 6;
@@ -922,12 +933,25 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
     int tr_row = 8 * (lane >> 4) + ((lane & 15) >> 1);
     int tr_half = lane & 1;
 
-    // staging: 128 threads per strip; thread covers 128 contiguous global
-    // bytes of one k-row (row tt>>1, byte half 128*(tt&1)).
+    // staging: 128 threads per strip; thread tt covers row tt>>1, chunk
+    // positions 2e + (tt&1) (interleaved halves: conflict-free b128
+    // writes, 32-B-contiguous paired global loads).
     int st_strip = tid >> 7;
     int tt = tid & 127;
     int st_row = tt >> 1;
-    int st_cq = (tt & 1) * 8;  // first of 8 consecutive 16-B chunks
+    int st_h = tt & 1;
+
+    // Per-lane LDS bases (byte offsets within lds), buf 0; buf 1 = +RS2_BUF.
+    // Reads: wave quadrant (wr, wc) takes chunks 8*wr+ta of strip 0 and
+    // 8*wc+tb of strip 1; every tr8 offset is base + 16*cc (+ hi half at
+    // +32*RS2_ROW), an immediate.
+    signed char* lds0 = &lds[0][0][0][0];
+    const signed char* rdI0 = lds0 + tr_row * RS2_ROW + 8 * tr_half
+                            + 16 * (8 * wr);
+    const signed char* rdJ0 = lds0 + RS2_STRIP + tr_row * RS2_ROW
+                            + 8 * tr_half + 16 * (8 * wc);
+    signed char* wr0 = lds0 + st_strip * RS2_STRIP + st_row * RS2_ROW
+                     + 16 * st_h;
 
     long total = 8 * ntiles * ((nbatch + 7) / 8);
     for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
@@ -942,8 +966,8 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
         long i0 = bi * 128, j0 = bj * 128;
         bool diag = bi == bj;
         // Diagonal tiles: the (wr=0,wc=1) quadrant lies entirely above the
-        // diagonal (skips MFMAs, still stages); in wr==wc quadrants the
-        // ta<tb MFMA tiles are above the diagonal and skip too.
+        // diagonal (skips MFMAs, still stages); wr==wc quadrants skip
+        // their ta<tb MFMA tiles via the reduced burst.
         bool skip_all = diag && wr < wc;
         bool diag_q = diag && wr == wc;
         v4i acc[8][8];
@@ -954,82 +978,103 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
 
         long base_col = st_strip ? j0 : i0;
         const long slab_step = 64 * lda * 2;
-        const signed char* src0 = ab + (long)st_row * lda * 2 +
-                                  base_col * 2 + 128 * (tt & 1);
         v4i stg[8];
-        const signed char* load_next = src0;
+        const signed char* load_next = ab + (long)st_row * lda * 2 +
+                                       base_col * 2 + 16 * st_h;
         auto load_slab = [&]() {
-            const v4i* pv = (const v4i*)__builtin_assume_aligned(
-                load_next, 16);
+            const signed char* p = load_next;
             load_next += slab_step;
 #pragma unroll
-            for (int e = 0; e < 8; ++e) stg[e] = pv[e];
+            for (int e = 0; e < 8; ++e)
+                stg[e] = *(const v4i*)__builtin_assume_aligned(p + 32 * e,
+                                                               16);
         };
-        auto write_slab = [&](int buf) {
-            int swz = st_row & 15;
-            signed char* base = &lds[buf][st_strip][st_row][0];
+        auto write_slab = [&](signed char* wbase) {
 #pragma unroll
             for (int e = 0; e < 8; ++e)
-                *(v4i*)(base + 16 * ((st_cq + e) ^ swz)) = stg[e];
+                *(v4i*)(wbase + 16 * (2 * e)) = stg[e];
         };
         auto frag = [&](const signed char* base, int cc) {
-            const signed char* p = base + tr_row * 256 +
-                16 * (cc ^ (tr_row & 15)) + 8 * tr_half;
-            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(base + 16 * cc));
             v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
-                (lds_v2i)(p + 32 * 256));
+                (lds_v2i)(base + 32 * RS2_ROW + 16 * cc));
             return v4i{lo[0], lo[1], hi[0], hi[1]};
         };
-        // 8x8 MFMA burst over the wave quadrant, fa two-rows-ahead
-        // pipelined (the rs sched-5 pattern at twice the width).
-        auto burst = [&](int buf) {
-            if (skip_all) return;
-            const signed char* bI = &lds[buf][0][0][0];
-            const signed char* bJ = &lds[buf][1][0][0];
-            v4i fb[8];
+        // Full-interior burst: all 16 fragments preloaded (64 VGPRs),
+        // then 64 branch-free MFMAs.
+        auto burst_full = [&](const signed char* bI, const signed char* bJ) {
+            v4i fa[8], fb[8];
 #pragma unroll
-            for (int tb = 0; tb < 8; ++tb)
-                fb[tb] = frag(bJ, 8 * wc + tb);
-            v4i fa0 = frag(bI, 8 * wr);
-            v4i fa1 = frag(bI, 8 * wr + 1);
-#pragma unroll
-            for (int ta = 0; ta < 8; ++ta) {
-                v4i fa2;
-                if (ta < 6) fa2 = frag(bI, 8 * wr + ta + 2);
-#pragma unroll
-                for (int tb = 0; tb < 8; ++tb) {
-                    if (diag_q && ta < tb) continue;
-                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
-                        fa0, fb[tb], acc[ta][tb], 0, 0, 0);
-                }
-                fa0 = fa1;
-                fa1 = fa2;
+            for (int u = 0; u < 8; ++u) {
+                fa[u] = frag(bI, u);
+                fb[u] = frag(bJ, u);
             }
+#pragma unroll
+            for (int ta = 0; ta < 8; ++ta)
+#pragma unroll
+                for (int tb = 0; tb < 8; ++tb)
+                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                        fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
         };
-
-        int nslab = (int)(k / 64);
-        load_slab();
-        write_slab(0);
-        if (nslab > 1) load_slab();
-        __syncthreads();
-        for (int s = 0; s < nslab; ++s) {
-            int buf = s & 1;
+        // Diagonal-quadrant burst: lower-triangle MFMA tiles only (36).
+        auto burst_diag = [&](const signed char* bI, const signed char* bJ) {
+            v4i fa[8], fb[8];
+#pragma unroll
+            for (int u = 0; u < 8; ++u) {
+                fa[u] = frag(bI, u);
+                fb[u] = frag(bJ, u);
+            }
+#pragma unroll
+            for (int ta = 0; ta < 8; ++ta)
+#pragma unroll
+                for (int tb = 0; tb < 8; ++tb)
+                    if (ta >= tb)
+                        acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                            fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+        };
+        auto step = [&](int buf, int s, int nslab) {
+            const signed char* bI = rdI0 + buf * RS2_BUF;
+            const signed char* bJ = rdJ0 + buf * RS2_BUF;
+            signed char* wb = wr0 + (buf ^ 1) * RS2_BUF;
             if (SCHED == 0) {
-                // early writes: they wait only the 2-slab-old stg loads
-                // (long landed) and drain across the MFMA burst.
                 if (s + 1 < nslab) {
-                    write_slab(buf ^ 1);
+                    write_slab(wb);
                     if (s + 2 < nslab) load_slab();
                 }
-                burst(buf);
+                if (skip_all) {
+                } else if (diag_q) {
+                    burst_diag(bI, bJ);
+                } else {
+                    burst_full(bI, bJ);
+                }
             } else {
-                burst(buf);
+                if (skip_all) {
+                } else if (diag_q) {
+                    burst_diag(bI, bJ);
+                } else {
+                    burst_full(bI, bJ);
+                }
                 if (s + 1 < nslab) {
-                    write_slab(buf ^ 1);
+                    write_slab(wb);
                     if (s + 2 < nslab) load_slab();
                 }
             }
             __syncthreads();
+        };
+
+        int nslab = (int)(k / 64);
+        load_slab();
+        write_slab(wr0);
+        if (nslab > 1) load_slab();
+        __syncthreads();
+        int s = 0;
+        while (s < nslab) {
+            step(0, s, nslab);
+            ++s;
+            if (s >= nslab) break;
+            step(1, s, nslab);
+            ++s;
         }
 #pragma unroll
         for (int ta = 0; ta < 8; ++ta) {
