@@ -1037,7 +1037,22 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
             const signed char* bI = rdI0 + buf * RS2_BUF;
             const signed char* bJ = rdJ0 + buf * RS2_BUF;
             signed char* wb = wr0 + (buf ^ 1) * RS2_BUF;
-            if (SCHED == 0) {
+            if (SCHED == 6) {
+                // DIAGNOSTIC (wrong results): compute-only ceiling — no
+                // staging, every slab re-reads the resident buffers.
+                if (skip_all) {
+                } else if (diag_q) {
+                    burst_diag(bI, bJ);
+                } else {
+                    burst_full(bI, bJ);
+                }
+            } else if (SCHED == 7) {
+                // DIAGNOSTIC (wrong results): staging-only floor.
+                if (s + 1 < nslab) {
+                    write_slab(wb);
+                    if (s + 2 < nslab) load_slab();
+                }
+            } else if (SCHED == 0) {
                 if (s + 1 < nslab) {
                     write_slab(wb);
                     if (s + 2 < nslab) load_slab();
@@ -1872,6 +1887,10 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 };
                 if (sched == 1)
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<1>);
+                else if (sched == 6)  // diagnostic: wrong results
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<6>);
+                else if (sched == 7)  // diagnostic: wrong results
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<7>);
                 else
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<0>);
                 BF_CHECK_HIP(hipGetLastError());
