@@ -158,11 +158,212 @@ def traffic_bytes_per_launch(workload_key):
     return None
 
 
+def run_c5(args):
+    """Config-5 pipeline benchmark (BASELINE configs[4]): 4-bit voltages
+    -> bf16-split-MFMA beamform (64 beams) -> fine-time FFT -> square-law
+    detect -> accumulate, run as a REAL bifrost_amd.pipeline over
+    cuda-space rings.  The input is HBM-resident: the source block fills
+    its ring spans device-to-device from a staged ci4 gulp, so no PCIe
+    traffic sits in the timed region.  A "step" is one gulp of
+    C5_TGULP time frames x 512 channels through the whole chain.
+    Prints ONE JSON line (metric distinct from the headline correlator
+    line; the driver's default invocation never reaches this mode)."""
+    import torch
+    import bifrost_amd as bf
+    from bifrost_amd import views
+    from bifrost_amd import device as bf_device
+    from bifrost_amd import memory as bf_memory
+    from bifrost_amd.linalg import LinAlg
+    from bifrost_amd.pipeline import Pipeline, SourceBlock, SinkBlock, \
+        TransformBlock
+
+    NBEAM, NCHAN_C5, NFINE = 64, 512, 64
+    TGULP = 256
+    n = N
+
+    bf_device.set_device(0)
+    torch.cuda.set_device(0)
+    bf_device.set_stream(torch.cuda.current_stream().cuda_stream)
+
+    rng = np.random.RandomState(1234)
+    re = rng.randint(-7, 8, size=(TGULP, NCHAN_C5, n))
+    im = rng.randint(-7, 8, size=(TGULP, NCHAN_C5, n))
+    packed = (((re & 0xF) << 4) | (im & 0xF)).astype(np.uint8)
+    dev_gulp = bf.asarray(bf.ndarray(packed.view(bf.DataType.ci4)
+                                     .reshape(TGULP, NCHAN_C5, n)),
+                          space="cuda")
+    w = (rng.standard_normal((NBEAM, NCHAN_C5, n, 2)).astype(np.float32)
+         .view(np.complex64).reshape(NBEAM, NCHAN_C5, n))
+
+    class _GulpReader(object):
+        def __init__(self, ngulp):
+            self.left = ngulp
+
+        def __enter__(self):
+            return self
+
+        def __exit__(self, *exc):
+            return False
+
+    class DeviceVoltageSource(SourceBlock):
+        """HBM-resident ci4 source: D2D copy into cuda-space ring spans."""
+
+        def __init__(self, ngulp, **kw):
+            super(DeviceVoltageSource, self).__init__(
+                ["c5"], TGULP, space="cuda", **kw)
+            self.ngulp = ngulp
+
+        def create_reader(self, sourcename):
+            return _GulpReader(self.ngulp)
+
+        def on_sequence(self, reader, sourcename):
+            return [{
+                "name": "c5-voltages",
+                "time_tag": 0,
+                "_tensor": {
+                    "dtype": "ci4",
+                    "shape": [-1, NCHAN_C5, n],
+                    "labels": ["time", "freq", "stand_pol"],
+                    "scales": [[0, 1]] * 3,
+                    "units": [None] * 3,
+                },
+                "gulp_nframe": TGULP,
+            }]
+
+        def on_data(self, reader, ospans):
+            if reader.left <= 0:
+                return [0]
+            reader.left -= 1
+            bf_memory.memcpy(ospans[0].data[:TGULP], dev_gulp)
+            return [TGULP]
+
+    class BeamformBlock(TransformBlock):
+        """W[b,c,n] cf32 x X[t,c,n] ci4 -> Y[t,c,b] cf32 (the bf16-split
+        MFMA beamformer; same block as tests/test_pipeline_c5_gpu.py)."""
+
+        def __init__(self, iring, weights, **kw):
+            super(BeamformBlock, self).__init__(iring, **kw)
+            self.weights_host = weights
+            self.linalg = LinAlg()
+            self.w = None
+            self.scratch = None
+
+        def define_valid_input_spaces(self):
+            return ("cuda",)
+
+        def on_sequence(self, iseq):
+            from copy import deepcopy
+            ohdr = deepcopy(iseq.header)
+            t = ohdr["_tensor"]
+            nbeam = self.weights_host.shape[0]
+            t["dtype"] = "cf32"
+            t["shape"] = [-1, t["shape"][1], nbeam]
+            t["labels"] = ["time", "freq", "beam"]
+            t["scales"] = [t["scales"][0], t["scales"][1], None]
+            t["units"] = [t["units"][0], t["units"][1], None]
+            self.w = bf.asarray(self.weights_host, space="cuda")
+            self.scratch = None
+            return ohdr
+
+        def on_data(self, ispan, ospan):
+            idata = ispan.data
+            odata = ospan.data
+            T = idata.shape[0]
+            nchan = idata.shape[1]
+            nbeam = self.w.shape[0]
+            if self.scratch is None or self.scratch.shape[2] != T:
+                self.scratch = bf.ndarray(shape=(nchan, nbeam, T),
+                                          dtype="cf32", space="cuda")
+            self.linalg.matmul(1, self.w.transpose(1, 0, 2),
+                               idata.transpose(1, 2, 0), 0, self.scratch)
+            bf.transpose(odata, self.scratch, (2, 0, 1))
+
+    class DrainBlock(SinkBlock):
+        def __init__(self, iring, **kw):
+            super(DrainBlock, self).__init__(iring, **kw)
+            self.nframe = 0
+
+        def on_sequence(self, iseq):
+            pass
+
+        def on_data(self, ispan):
+            self.nframe += ispan.nframe
+
+    def run_pipeline(ngulp):
+        drain = []
+        with Pipeline() as pipe:
+            src = DeviceVoltageSource(ngulp)
+            beam = BeamformBlock(src, w)
+            fine = views.split_axis(beam, 0, NFINE, label="fine_time")
+            spec = bf.blocks.fft(fine, axes="fine_time")
+            pwr = bf.blocks.detect(spec, mode="scalar")
+            acc = bf.blocks.accumulate(pwr, TGULP // NFINE)
+            drain.append(DrainBlock(acc))
+            t0 = time.perf_counter()
+            pipe.run()
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+        return dt, drain[0].nframe
+
+    run_pipeline(max(2, args.warmup))           # spin-up: JIT, FFT plans
+    dt, nspec = run_pipeline(args.steps)
+    samples = args.steps * TGULP * NCHAN_C5      # (time, chan) samples
+    value = samples / dt / 1e9
+    # The dominant kernel is the bf16-split MFMA beamformer: flops =
+    # 8 real ops x NBEAM x n complex MACs per (t,chan) sample, against
+    # the ~1.25 PF split-bf16 effective ceiling (DESIGN.md §3; per-kernel
+    # event evidence in profiles/round2_c5.md).  Wall-derived achieved is
+    # a LOWER bound (the wall includes fft/detect/accumulate).
+    flops = samples * NBEAM * n * 8.0
+    achieved_tf = flops / dt / 1e12
+    result = {
+        "metric": "C5 beamform-pipeline samples processed",
+        "value": round(value, 4),
+        "unit": "Gsamp/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(dt / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "ci4",
+        "data": "synthetic",
+        "config": {
+            "workload": "c5_beamform_fft_detect_b%d_c%d_t%d_f%d"
+                        % (NBEAM, NCHAN_C5, TGULP, NFINE),
+            "nstand": NSTAND,
+            "npol": NPOL,
+            "nbeam": NBEAM,
+            "nchan": NCHAN_C5,
+            "tgulp": TGULP,
+            "nfine": NFINE,
+            "parallelism": "single-GPU pipeline",
+        },
+        "roofline": {
+            "bound": "mfma",
+            "achieved": round(achieved_tf, 1),
+            "peak": 1250.0,
+            "unit": "TFLOP/s",
+            "frac": round(achieved_tf / 1250.0, 4),
+            "traffic": None,
+            "note": "wall-derived lower bound over the whole pipeline; "
+                    "per-kernel split in profiles/round2_c5.md",
+        },
+        "drained_frames": nspec,
+    }
+    print(json.dumps(result))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--mode", choices=["xcorr", "c5"], default="xcorr",
+                    help="xcorr = the headline correlator benchmark "
+                         "(driver contract); c5 = the config-5 "
+                         "beamform+FFT+detect pipeline line")
     ap.add_argument("--time-split", action="store_true",
                     help="split one integration in time across ranks and "
                          "all-reduce visibilities over RCCL")
@@ -170,6 +371,10 @@ def main():
     ap.add_argument("--nchan", type=int, default=NCHAN_PER_GPU)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
+
+    if args.mode == "c5":
+        run_c5(args)
+        return
 
     import torch
 

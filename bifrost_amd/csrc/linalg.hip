@@ -1001,15 +1001,20 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
                 (lds_v2i)(base + 32 * RS2_ROW + 16 * cc));
             return v4i{lo[0], lo[1], hi[0], hi[1]};
         };
-        // Full-interior burst: all 16 fragments preloaded (64 VGPRs),
-        // then 64 branch-free MFMAs.
+        // Full-interior burst: ALL 16 fragments issue before the first
+        // MFMA (fb then fa, so rows consume reads in completion order)
+        // and a sched_barrier pins the order — without it the scheduler
+        // interleaves reads into the burst and each lgkm group exposes a
+        // ~50-cycle park that occupancy 1 cannot hide (measured: the
+        // compute-only ablation sat 3.3x above the 16.3 cyc/MFMA pipe
+        // rate probed in csrc/probe_mfma_rate.hip).
         auto burst_full = [&](const signed char* bI, const signed char* bJ) {
             v4i fa[8], fb[8];
 #pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                fa[u] = frag(bI, u);
-                fb[u] = frag(bJ, u);
-            }
+            for (int u = 0; u < 8; ++u) fb[u] = frag(bJ, u);
+#pragma unroll
+            for (int u = 0; u < 8; ++u) fa[u] = frag(bI, u);
+            __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
             for (int ta = 0; ta < 8; ++ta)
 #pragma unroll
@@ -1021,10 +1026,10 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
         auto burst_diag = [&](const signed char* bI, const signed char* bJ) {
             v4i fa[8], fb[8];
 #pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                fa[u] = frag(bI, u);
-                fb[u] = frag(bJ, u);
-            }
+            for (int u = 0; u < 8; ++u) fb[u] = frag(bJ, u);
+#pragma unroll
+            for (int u = 0; u < 8; ++u) fa[u] = frag(bI, u);
+            __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
             for (int ta = 0; ta < 8; ++ta)
 #pragma unroll

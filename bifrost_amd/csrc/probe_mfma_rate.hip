@@ -1,0 +1,141 @@
+// Microprobe: back-to-back v_mfma_i32_16x16x64_i8 issue rate by
+// occupancy and accumulator count on gfx950 (round-2: the rs2 kernel's
+// compute-only ablation sits 3.2x above the 20.4 cyc/instr rate derived
+// from the 3944-TOPS peak — this pins the real per-wave issue rate at
+// 1 wave/SIMD with a 256-register accumulator file, vs 4 waves/SIMD).
+//
+// Build: hipcc -O3 --offload-arch=gfx950 probe_mfma_rate.hip -o probe_mfma_rate
+// Run on the GPU box: ./probe_mfma_rate
+
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+
+typedef int v4i __attribute__((ext_vector_type(4)));
+
+// NACC accumulators round-robin, ITER*NACC MFMAs per wave.
+template <int NACC, int WPE>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(WPE)))
+void mfma_rate_kernel(const int* __restrict__ in, long* __restrict__ cyc,
+                      int* __restrict__ sink, int iter) {
+    v4i acc[NACC];
+    v4i a, b;
+    int lane = threadIdx.x & 63;
+    a = v4i{in[lane], in[lane + 64], in[lane + 128], in[lane + 192]};
+    b = v4i{in[lane + 1], in[lane + 65], in[lane + 129], in[lane + 193]};
+#pragma unroll
+    for (int i = 0; i < NACC; ++i)
+        acc[i] = v4i{in[i], 0, 0, 0};
+    __syncthreads();
+    long t0 = __builtin_amdgcn_s_memtime();
+    for (int it = 0; it < iter; ++it) {
+#pragma unroll
+        for (int i = 0; i < NACC; ++i)
+            acc[i] = __builtin_amdgcn_mfma_i32_16x16x64_i8(a, b, acc[i],
+                                                           0, 0, 0);
+    }
+    long t1 = __builtin_amdgcn_s_memtime();
+    int s = 0;
+#pragma unroll
+    for (int i = 0; i < NACC; ++i) s += acc[i][0] + acc[i][3];
+    if (threadIdx.x == 0) {
+        sink[blockIdx.x] = s;
+        cyc[blockIdx.x] = t1 - t0;
+    }
+}
+
+// Same, with 2 ds_read_b64_tr8 per MFMA interleaved (the rs2 fragment
+// read mix) to price the co-issue.
+typedef int v2i __attribute__((ext_vector_type(2)));
+typedef __attribute__((address_space(3))) v2i* lds_v2i;
+
+template <int NACC, int WPE>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(WPE)))
+void mfma_tr8_kernel(const int* __restrict__ in, long* __restrict__ cyc,
+                     int* __restrict__ sink, int iter) {
+    __shared__ signed char lds[64][272];
+    v4i acc[NACC];
+    v4i a, b;
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    a = v4i{in[lane], in[lane + 64], in[lane + 128], in[lane + 192]};
+    b = v4i{in[lane + 1], in[lane + 65], in[lane + 129], in[lane + 193]};
+#pragma unroll
+    for (int i = 0; i < NACC; ++i)
+        acc[i] = v4i{in[i], 0, 0, 0};
+    for (int i = tid; i < 64 * 272 / 4; i += 256)
+        ((__attribute__((address_space(3))) int*)&lds[0][0])[i] = in[i & 255];
+    __syncthreads();
+    int tr_row = 8 * (lane >> 4) + ((lane & 15) >> 1);
+    int tr_half = lane & 1;
+    const signed char* base =
+        (const signed char*)&lds[0][0] + tr_row * 272 + 8 * tr_half;
+    long t0 = __builtin_amdgcn_s_memtime();
+    for (int it = 0; it < iter; ++it) {
+#pragma unroll
+        for (int i = 0; i < NACC; ++i) {
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(base + 16 * (i & 15)));
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(base + 32 * 272 + 16 * (i & 15)));
+            v4i f = v4i{lo[0], lo[1], hi[0], hi[1]};
+            acc[i] = __builtin_amdgcn_mfma_i32_16x16x64_i8(a, f, acc[i],
+                                                           0, 0, 0);
+        }
+    }
+    long t1 = __builtin_amdgcn_s_memtime();
+    int s = 0;
+#pragma unroll
+    for (int i = 0; i < NACC; ++i) s += acc[i][0] + acc[i][3];
+    if (threadIdx.x == 0) {
+        sink[blockIdx.x] = s;
+        cyc[blockIdx.x] = t1 - t0;
+    }
+}
+
+template <typename K>
+void run(const char* name, K kern, int blocks, int iter, int nacc) {
+    int* in;
+    long* cyc;
+    int* sink;
+    (void)hipMalloc(&in, 4096);
+    (void)hipMalloc(&cyc, blocks * sizeof(long));
+    (void)hipMalloc(&sink, blocks * sizeof(int));
+    (void)hipMemset(in, 1, 4096);
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, 0, in, cyc, sink,
+                       iter);
+    (void)hipDeviceSynchronize();
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, 0, in, cyc, sink,
+                       iter);
+    (void)hipDeviceSynchronize();
+    long* h = new long[blocks];
+    (void)hipMemcpy(h, cyc, blocks * sizeof(long), hipMemcpyDeviceToHost);
+    long mx = 0, mn = (long)1e18;
+    for (int i = 0; i < blocks; ++i) {
+        if (h[i] > mx) mx = h[i];
+        if (h[i] < mn) mn = h[i];
+    }
+    double per = (double)mx / ((double)iter * nacc);
+    printf("%-28s blocks=%4d nacc=%3d  cyc/MFMA max %.2f min %.2f\n", name,
+           blocks, nacc, per, (double)mn / ((double)iter * nacc));
+    delete[] h;
+    (void)hipFree(in); (void)hipFree(cyc); (void)hipFree(sink);
+}
+
+int main() {
+    int iter = 2000;
+    // pure MFMA chains
+    run("occ1 nacc64 (rs2 shape)", mfma_rate_kernel<64, 1>, 256, iter, 64);
+    run("occ1 nacc16", mfma_rate_kernel<16, 1>, 256, iter, 16);
+    run("occ2 nacc32", mfma_rate_kernel<32, 2>, 512, iter, 32);
+    run("occ2 nacc16", mfma_rate_kernel<16, 2>, 512, iter, 16);
+    run("occ4 nacc16 (rs shape)", mfma_rate_kernel<16, 4>, 1024, iter, 16);
+    run("occ4 nacc8", mfma_rate_kernel<8, 4>, 1024, iter, 8);
+    // MFMA + 2x tr8 per MFMA (the fragment-read mix)
+    run("tr8 occ1 nacc64", mfma_tr8_kernel<64, 1>, 256, iter, 64);
+    run("tr8 occ2 nacc32", mfma_tr8_kernel<32, 2>, 512, iter, 32);
+    run("tr8 occ4 nacc16", mfma_tr8_kernel<16, 4>, 1024, iter, 16);
+    return 0;
+}
