@@ -1038,11 +1038,48 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
                         acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
                             fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
         };
+        // Diagnostic fixed fragments (SCHED 9/11): read once per tile so
+        // the k-loop's MFMA stream runs with the real register pattern
+        // but zero LDS reads.
+        v4i ffa[8], ffb[8];
+        if (SCHED == 9 || SCHED == 11) {
+#pragma unroll
+            for (int u = 0; u < 8; ++u) {
+                ffb[u] = frag(rdJ0, u);
+                ffa[u] = frag(rdI0, u);
+            }
+        }
+        auto burst_fixed = [&]() {
+#pragma unroll
+            for (int ta = 0; ta < 8; ++ta)
+#pragma unroll
+                for (int tb = 0; tb < 8; ++tb)
+                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                        ffa[ta], ffb[tb], acc[ta][tb], 0, 0, 0);
+        };
         auto step = [&](int buf, int s, int nslab) {
             const signed char* bI = rdI0 + buf * RS2_BUF;
             const signed char* bJ = rdJ0 + buf * RS2_BUF;
             signed char* wb = wr0 + (buf ^ 1) * RS2_BUF;
-            if (SCHED == 6) {
+            if (SCHED == 8) {
+                // DIAGNOSTIC: compute-only, NO per-slab barrier.
+                if (skip_all) {
+                } else if (diag_q) {
+                    burst_diag(bI, bJ);
+                } else {
+                    burst_full(bI, bJ);
+                }
+                return;
+            } else if (SCHED == 9) {
+                // DIAGNOSTIC: fixed-fragment MFMA stream + barrier
+                // (no LDS reads in the loop).
+                if (!skip_all) burst_fixed();
+            } else if (SCHED == 11) {
+                // DIAGNOSTIC: fixed-fragment MFMA stream, no barrier —
+                // should reproduce the probe's 16.3 cyc/MFMA floor.
+                if (!skip_all) burst_fixed();
+                return;
+            } else if (SCHED == 6) {
                 // DIAGNOSTIC (wrong results): compute-only ceiling — no
                 // staging, every slab re-reads the resident buffers.
                 if (skip_all) {
@@ -1539,9 +1576,16 @@ typedef bf16_t v8bf __attribute__((ext_vector_type(8)));
 typedef float v4f __attribute__((ext_vector_type(4)));
 
 // XT: 0 = ci8 X, 1 = ci4 X; WT: 0 = cf32 W (hi/lo bf16 RNE split),
-// 1 = ci16 W (exact high-byte/low-byte split)
-template <int NBT, int XT = 0, int WT = 0, int JT = 2>
-__global__ __launch_bounds__(256) void beamform_mfma_kernel(
+// 1 = ci16 W (exact high-byte/low-byte split).
+// PRE: 1 = register-prefetched staging (round-1 form: wpre/xpre live
+// across the MFMA burst -> 212 VGPR, 2 waves/SIMD); 0 = fetch at slab
+// start (wpre/xpre short-lived, allocator reuses them -> targets
+// 3 waves/SIMD; staging latency covered by the extra wave instead of
+// the prefetch distance).  Round-2 A/B: BIFROST_BEAM=mfma3.
+template <int NBT, int XT = 0, int WT = 0, int JT = 2, int PRE = 1>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(PRE ? 2 : 3)))
+void beamform_mfma_kernel(
     long nn, long k, long nbatch, float alpha, const void* __restrict__ w_,
     long ldw, long w_b, const signed char* __restrict__ x, long ldx,
     long x_b, float beta, f2* __restrict__ c, long c_row, long c_b,
@@ -1666,11 +1710,12 @@ __global__ __launch_bounds__(256) void beamform_mfma_kernel(
                     }
                 }
             };
-            fetch(0);
+            if (PRE) fetch(0);
             for (long k0 = 0; k0 < k; k0 += 64) {
+                if (!PRE) fetch(k0);
                 commit();
                 __syncthreads();
-                if (k0 + 64 < k) fetch(k0 + 64);
+                if (PRE && k0 + 64 < k) fetch(k0 + 64);
                 for (int kc = 0; kc < 2; ++kc) {
                     int kof = 32 * kc + 8 * kblk;
                     int sw = 8 * ((kof >> 3) ^ (row16 & 7));
@@ -1896,6 +1941,12 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<6>);
                 else if (sched == 7)  // diagnostic: wrong results
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<7>);
+                else if (sched == 8)  // diagnostic: wrong results
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<8>);
+                else if (sched == 9)  // diagnostic: wrong results
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<9>);
+                else if (sched == 11)  // diagnostic: wrong results
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<11>);
                 else
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<0>);
                 BF_CHECK_HIP(hipGetLastError());
@@ -2063,11 +2114,24 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
             long max_chunk = ck ? atol(ck) : 64;
             for (long i0 = 0; i0 < m;) {
                 long chunk = std::min<long>(max_chunk, m - i0);
+// BIFROST_BEAM=mfma3 selects the no-prefetch 3-wave/SIMD variant
+// (PRE=0); default stays the round-1 prefetch form until the same-box
+// A/B decides.
 #define BEAM_MFMA_ONE(NBT, XTV, WTV)                                          \
-    hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV>), mgrid,          \
-                       dim3(256), 0, stream, nn, k, nbatch, (float)alpha,     \
-                       a, a_i, a_b, (const signed char*)b, b_j, b_b,          \
-                       (float)beta, (f2*)c, c_row, c_b, i0)
+    do {                                                                      \
+        if (beam_sel && strcmp(beam_sel, "mfma3") == 0)                       \
+            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV, 2, 0>),   \
+                               mgrid, dim3(256), 0, stream, nn, k, nbatch,    \
+                               (float)alpha, a, a_i, a_b,                     \
+                               (const signed char*)b, b_j, b_b, (float)beta,  \
+                               (f2*)c, c_row, c_b, i0);                       \
+        else                                                                  \
+            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV>), mgrid,  \
+                               dim3(256), 0, stream, nn, k, nbatch,           \
+                               (float)alpha, a, a_i, a_b,                     \
+                               (const signed char*)b, b_j, b_b, (float)beta,  \
+                               (f2*)c, c_row, c_b, i0);                       \
+    } while (0)
 #define BEAM_MFMA_CASE(NBT)                                                   \
     do {                                                                      \
         if (x4 && w16) BEAM_MFMA_ONE(NBT, 1, 1);                              \
