@@ -2114,23 +2114,27 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
             long max_chunk = ck ? atol(ck) : 64;
             for (long i0 = 0; i0 < m;) {
                 long chunk = std::min<long>(max_chunk, m - i0);
-// BIFROST_BEAM=mfma3 selects the no-prefetch 3-wave/SIMD variant
-// (PRE=0); default stays the round-1 prefetch form until the same-box
-// A/B decides.
+// BIFROST_BEAM=mfma3 selects the 3-wave/SIMD variant: JT=1 (48 KB LDS
+// -> 3 workgroups/CU; the JT=2 form's 64 KB caps occupancy at 2
+// regardless of registers) and PRE=0 (no cross-slab register prefetch;
+// the third wave covers staging latency instead).  Default stays the
+// round-1 prefetch form until the same-box A/B decides.
 #define BEAM_MFMA_ONE(NBT, XTV, WTV)                                          \
     do {                                                                      \
-        if (beam_sel && strcmp(beam_sel, "mfma3") == 0)                       \
-            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV, 2, 0>),   \
-                               mgrid, dim3(256), 0, stream, nn, k, nbatch,    \
+        if (beam_sel && strcmp(beam_sel, "mfma3") == 0) {                     \
+            dim3 mgrid3(cap_grid(nn / 64, 4096), cap_grid(nbatch, 65535));    \
+            hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV, 1, 0>),   \
+                               mgrid3, dim3(256), 0, stream, nn, k, nbatch,   \
                                (float)alpha, a, a_i, a_b,                     \
                                (const signed char*)b, b_j, b_b, (float)beta,  \
                                (f2*)c, c_row, c_b, i0);                       \
-        else                                                                  \
+        } else {                                                              \
             hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV>), mgrid,  \
                                dim3(256), 0, stream, nn, k, nbatch,           \
                                (float)alpha, a, a_i, a_b,                     \
                                (const signed char*)b, b_j, b_b, (float)beta,  \
                                (f2*)c, c_row, c_b, i0);                       \
+        }                                                                     \
     } while (0)
 #define BEAM_MFMA_CASE(NBT)                                                   \
     do {                                                                      \

@@ -45,6 +45,142 @@ void mfma_rate_kernel(const int* __restrict__ in, long* __restrict__ cyc,
     }
 }
 
+// Kernel-pattern variant: 8x8 DIVERSE operand pairs (fa[ta], fb[tb])
+// over 64 accumulators — exactly the rs2 burst's register pattern.
+// Prices MFMA operand-diversity vs the fixed-operand chain above.
+template <int WPE>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(WPE)))
+void mfma_pairs_kernel(const int* __restrict__ in, long* __restrict__ cyc,
+                       int* __restrict__ sink, int iter) {
+    v4i acc[8][8];
+    v4i fa[8], fb[8];
+    int lane = threadIdx.x & 63;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+        fa[u] = v4i{in[lane + u], in[lane + u + 64], in[lane + u + 128],
+                    in[lane + u + 192]};
+        fb[u] = v4i{in[lane + u + 1], in[lane + u + 65],
+                    in[lane + u + 129], in[lane + u + 193]};
+    }
+#pragma unroll
+    for (int x = 0; x < 8; ++x)
+#pragma unroll
+        for (int y = 0; y < 8; ++y) acc[x][y] = v4i{in[x], 0, 0, in[y]};
+    __syncthreads();
+    long t0 = __builtin_amdgcn_s_memtime();
+    for (int it = 0; it < iter; ++it) {
+#pragma unroll
+        for (int ta = 0; ta < 8; ++ta)
+#pragma unroll
+            for (int tb = 0; tb < 8; ++tb)
+                acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                    fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+    }
+    long t1 = __builtin_amdgcn_s_memtime();
+    int s = 0;
+#pragma unroll
+    for (int x = 0; x < 8; ++x)
+#pragma unroll
+        for (int y = 0; y < 8; ++y) s += acc[x][y][0] + acc[x][y][3];
+    if (threadIdx.x == 0) {
+        sink[blockIdx.x] = s;
+        cyc[blockIdx.x] = t1 - t0;
+    }
+}
+
+// rsm geometry: fa[8] x fb[4] -> 32 accumulators (128 regs) at
+// 2 waves/SIMD — do co-resident waves hide the operand-switch stalls?
+template <int WPE>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(WPE)))
+void mfma_pairs2_kernel(const int* __restrict__ in, long* __restrict__ cyc,
+                        int* __restrict__ sink, int iter) {
+    v4i acc[8][4];
+    v4i fa[8], fb[4];
+    int lane = threadIdx.x & 63;
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+        fa[u] = v4i{in[lane + u], in[lane + u + 64], in[lane + u + 128],
+                    in[lane + u + 192]};
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+        fb[u] = v4i{in[lane + u + 1], in[lane + u + 65],
+                    in[lane + u + 129], in[lane + u + 193]};
+#pragma unroll
+    for (int x = 0; x < 8; ++x)
+#pragma unroll
+        for (int y = 0; y < 4; ++y) acc[x][y] = v4i{in[x], 0, 0, in[y]};
+    __syncthreads();
+    long t0 = __builtin_amdgcn_s_memtime();
+    for (int it = 0; it < iter; ++it) {
+#pragma unroll
+        for (int ta = 0; ta < 8; ++ta)
+#pragma unroll
+            for (int tb = 0; tb < 4; ++tb)
+                acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                    fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+    }
+    long t1 = __builtin_amdgcn_s_memtime();
+    int s = 0;
+#pragma unroll
+    for (int x = 0; x < 8; ++x)
+#pragma unroll
+        for (int y = 0; y < 4; ++y) s += acc[x][y][0] + acc[x][y][3];
+    if (threadIdx.x == 0) {
+        sink[blockIdx.x] = s;
+        cyc[blockIdx.x] = t1 - t0;
+    }
+}
+
+// v_mfma_i32_32x32x32_i8: 4x the ops per instruction, so any per-
+// instruction operand-switch stall amortizes 4x.  Fixed vs diverse.
+typedef int v16i __attribute__((ext_vector_type(16)));
+
+template <int WPE, int DIVERSE>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(WPE)))
+void mfma32_kernel(const int* __restrict__ in, long* __restrict__ cyc,
+                   int* __restrict__ sink, int iter) {
+    v16i acc[4][4];
+    v4i fa[4], fb[4];
+    int lane = threadIdx.x & 63;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+        fa[u] = v4i{in[lane + u], in[lane + u + 64], in[lane + u + 128],
+                    in[lane + u + 192]};
+        fb[u] = v4i{in[lane + u + 1], in[lane + u + 65],
+                    in[lane + u + 129], in[lane + u + 193]};
+    }
+#pragma unroll
+    for (int x = 0; x < 4; ++x)
+#pragma unroll
+        for (int y = 0; y < 4; ++y)
+#pragma unroll
+            for (int e = 0; e < 16; ++e) acc[x][y][e] = in[x + y + e];
+    __syncthreads();
+    long t0 = __builtin_amdgcn_s_memtime();
+    for (int it = 0; it < iter; ++it) {
+#pragma unroll
+        for (int ta = 0; ta < 4; ++ta)
+#pragma unroll
+            for (int tb = 0; tb < 4; ++tb)
+                acc[ta][tb] = __builtin_amdgcn_mfma_i32_32x32x32_i8(
+                    DIVERSE ? fa[ta] : fa[0], DIVERSE ? fb[tb] : fb[0],
+                    acc[ta][tb], 0, 0, 0);
+    }
+    long t1 = __builtin_amdgcn_s_memtime();
+    int s = 0;
+#pragma unroll
+    for (int x = 0; x < 4; ++x)
+#pragma unroll
+        for (int y = 0; y < 4; ++y) s += acc[x][y][0] + acc[x][y][15];
+    if (threadIdx.x == 0) {
+        sink[blockIdx.x] = s;
+        cyc[blockIdx.x] = t1 - t0;
+    }
+}
+
 // Same, with 2 ds_read_b64_tr8 per MFMA interleaved (the rs2 fragment
 // read mix) to price the co-issue.
 typedef int v2i __attribute__((ext_vector_type(2)));
@@ -133,6 +269,14 @@ int main() {
     run("occ2 nacc16", mfma_rate_kernel<16, 2>, 512, iter, 16);
     run("occ4 nacc16 (rs shape)", mfma_rate_kernel<16, 4>, 1024, iter, 16);
     run("occ4 nacc8", mfma_rate_kernel<8, 4>, 1024, iter, 8);
+    // diverse 8x8 operand pairs (the rs2 burst register pattern)
+    run("pairs occ1 (rs2 burst)", mfma_pairs_kernel<1>, 256, iter, 64);
+    run("pairs2 occ2 (rsm 8x4)", mfma_pairs2_kernel<2>, 512, iter, 32);
+    run("pairs2 occ1 (8x4)", mfma_pairs2_kernel<1>, 256, iter, 32);
+    // 32x32x32 i8 (16 i32 acc per instr; cyc shown PER 16x16x64-EQUIV,
+    // i.e. raw cyc/instr divided by 2 — compare directly with rows above)
+    run("mfma32 occ1 fixed", mfma32_kernel<1, 0>, 256, iter, 32);
+    run("mfma32 occ1 diverse 4x4", mfma32_kernel<1, 1>, 256, iter, 32);
     // MFMA + 2x tr8 per MFMA (the fragment-read mix)
     run("tr8 occ1 nacc64", mfma_tr8_kernel<64, 1>, 256, iter, 64);
     run("tr8 occ2 nacc32", mfma_tr8_kernel<32, 2>, 512, iter, 32);
