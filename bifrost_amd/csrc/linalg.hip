@@ -230,6 +230,7 @@ __global__ __launch_bounds__(256) void herk_generic_kernel(long n, long k, long 
 
 typedef int v4i __attribute__((ext_vector_type(4)));
 typedef int v2i __attribute__((ext_vector_type(2)));
+typedef int v16i __attribute__((ext_vector_type(16)));
 typedef __attribute__((address_space(3))) v2i* lds_v2i;
 typedef __attribute__((address_space(3))) unsigned* lds_u32;
 typedef const __attribute__((address_space(1))) unsigned* glob_u32;
@@ -1162,6 +1163,225 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
     }
 }
 
+/* ---- rs3: big-tile cherk on v_mfma_i32_32x32x32_i8 (round-2 winner) ---- */
+// Same 128x128-complex workgroup tile, staging, and padded [64][272]
+// strips as rs2, but the burst runs on the 32x32x32 i8 matrix op:
+// each wave's 64x64-complex quadrant is a 4x4 grid of 32x32-BYTE tiles,
+// 2 k-halves of 16 MFMAs each per 64-k slab = 32 instructions carrying
+// the same math as rs2's 64.  Probe evidence (csrc/probe_mfma_rate.hip,
+// gpurun_out/probe_mfma_rate3.log): at 1 wave/SIMD the 16x16x64 op pays
+// a ~12 cyc/instr operand-switch stall with DIVERSE (fa,fb) pairs (27.9
+// vs 16.3 cyc back-to-back), while 32x32x32 runs at full rate (16.0
+// cyc/16x16x64-equivalent) even fully diverse - the stall amortizes
+// over 4x the ops.  Layout verified on hardware (csrc/probe_mfma32.hip):
+//   A: lane l -> A[row=l&31][k=16*(l>>5)+e];  B symmetric;
+//   D: lane l, reg r -> D[row=8*(r>>2)+4*(l>>5)+(r&3)][col=l&31].
+// tr8 gather for that layout (derived from the probed gather model
+// received[l][j] = mem[addr[(l&0x30)|(2j)|((l>>3)&1)] + (l&7)]):
+// lane m supplies  row32 = 16*(m>>5) + ((m>>1)&7),
+//                  colb32 = 8*((m&1) + 2*((m>>4)&1));
+// a fragment (k-half h, tr8-half q, 32-byte tile cc) reads at
+//   strip + (32h + 8q + row32)*272 + 32*cc + colb32
+// which is bank-conflict-free on the 272-B rows (banks 4r + {0,2,4,6}
+// all distinct per gather group).  Requirements as rs2.
+template <int SCHED>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(1)))
+void cherk_ci8_mfma32_rs3_kernel(long n, long k, long nbatch, float alpha,
+                                 const signed char* __restrict__ a, long lda,
+                                 long a_b, float beta, f2* __restrict__ c,
+                                 long c_row, long c_b, long ntiles) {
+    __shared__ signed char lds[2][2][64][RS2_ROW];
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;
+
+    // per-lane tr8 source role for the 32x32x32 fragment gather
+    int row32 = 16 * (lane >> 5) + ((lane >> 1) & 7);
+    int colb32 = 8 * ((lane & 1) + 2 * ((lane >> 4) & 1));
+
+    int st_strip = tid >> 7;
+    int tt = tid & 127;
+    int st_row = tt >> 1;
+    int st_h = tt & 1;
+
+    signed char* lds0 = &lds[0][0][0][0];
+    // read bases: + quadrant offset (128 B per wave row/col half)
+    const signed char* rdI0 = lds0 + row32 * RS2_ROW + colb32 + 128 * wr;
+    const signed char* rdJ0 = lds0 + RS2_STRIP + row32 * RS2_ROW + colb32
+                            + 128 * wc;
+    signed char* wr0 = lds0 + st_strip * RS2_STRIP + st_row * RS2_ROW
+                     + 16 * st_h;
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        long bi, bj;
+        lift_tri(t, bi, bj);
+        long i0 = bi * 128, j0 = bj * 128;
+        bool diag = bi == bj;
+        bool skip_all = diag && wr < wc;
+        bool diag_q = diag && wr == wc;
+        v16i acc[4][4];
+#pragma unroll
+        for (int x = 0; x < 4; ++x)
+#pragma unroll
+            for (int y = 0; y < 4; ++y) acc[x][y] = v16i{};
+
+        long base_col = st_strip ? j0 : i0;
+        const long slab_step = 64 * lda * 2;
+        v4i stg[8];
+        const signed char* load_next = ab + (long)st_row * lda * 2 +
+                                       base_col * 2 + 16 * st_h;
+        auto load_slab = [&]() {
+            const signed char* p = load_next;
+            load_next += slab_step;
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                stg[e] = *(const v4i*)__builtin_assume_aligned(p + 32 * e,
+                                                               16);
+        };
+        auto write_slab = [&](signed char* wbase) {
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                *(v4i*)(wbase + 16 * (2 * e)) = stg[e];
+        };
+        // fragment: k-half h (0/1), 32-byte tile cc within the quadrant
+        auto frag32 = [&](const signed char* base, int h, int cc) {
+            const signed char* p = base + (32 * h) * RS2_ROW + 32 * cc;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 8 * RS2_ROW));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto burst_full = [&](const signed char* bI, const signed char* bJ) {
+            v4i fa[2][4], fb[2][4];
+#pragma unroll
+            for (int h = 0; h < 2; ++h)
+#pragma unroll
+                for (int u = 0; u < 4; ++u) {
+                    fb[h][u] = frag32(bJ, h, u);
+                    fa[h][u] = frag32(bI, h, u);
+                }
+            __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+            for (int h = 0; h < 2; ++h)
+#pragma unroll
+                for (int ta = 0; ta < 4; ++ta)
+#pragma unroll
+                    for (int tb = 0; tb < 4; ++tb)
+                        acc[ta][tb] = __builtin_amdgcn_mfma_i32_32x32x32_i8(
+                            fa[h][ta], fb[h][tb], acc[ta][tb], 0, 0, 0);
+        };
+        auto burst_diag = [&](const signed char* bI, const signed char* bJ) {
+            v4i fa[2][4], fb[2][4];
+#pragma unroll
+            for (int h = 0; h < 2; ++h)
+#pragma unroll
+                for (int u = 0; u < 4; ++u) {
+                    fb[h][u] = frag32(bJ, h, u);
+                    fa[h][u] = frag32(bI, h, u);
+                }
+            __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+            for (int h = 0; h < 2; ++h)
+#pragma unroll
+                for (int ta = 0; ta < 4; ++ta)
+#pragma unroll
+                    for (int tb = 0; tb < 4; ++tb)
+                        if (ta >= tb)
+                            acc[ta][tb] =
+                                __builtin_amdgcn_mfma_i32_32x32x32_i8(
+                                    fa[h][ta], fb[h][tb], acc[ta][tb],
+                                    0, 0, 0);
+        };
+        auto step = [&](int buf, int s, int nslab) {
+            const signed char* bI = rdI0 + buf * RS2_BUF;
+            const signed char* bJ = rdJ0 + buf * RS2_BUF;
+            signed char* wb = wr0 + (buf ^ 1) * RS2_BUF;
+            if (SCHED == 0) {
+                if (s + 1 < nslab) {
+                    write_slab(wb);
+                    if (s + 2 < nslab) load_slab();
+                }
+                if (skip_all) {
+                } else if (diag_q) {
+                    burst_diag(bI, bJ);
+                } else {
+                    burst_full(bI, bJ);
+                }
+            } else {
+                if (skip_all) {
+                } else if (diag_q) {
+                    burst_diag(bI, bJ);
+                } else {
+                    burst_full(bI, bJ);
+                }
+                if (s + 1 < nslab) {
+                    write_slab(wb);
+                    if (s + 2 < nslab) load_slab();
+                }
+            }
+            __syncthreads();
+        };
+
+        int nslab = (int)(k / 64);
+        load_slab();
+        write_slab(wr0);
+        if (nslab > 1) load_slab();
+        __syncthreads();
+        int s = 0;
+        while (s < nslab) {
+            step(0, s, nslab);
+            ++s;
+            if (s >= nslab) break;
+            step(1, s, nslab);
+            ++s;
+        }
+        // epilogue: D[row=8*(g)+4*(lane>>5)+(2p+c)][col=lane&31] per tile;
+        // complex combine pairs byte rows intra-lane (regs 4g+2p,+1) and
+        // byte cols across lane^1.
+#pragma unroll
+        for (int ta = 0; ta < 4; ++ta) {
+#pragma unroll
+            for (int tb = 0; tb < 4; ++tb) {
+                if (skip_all || (diag_q && ta < tb)) continue;
+                long ci_base = i0 + 64 * wr + 16 * ta;
+                long cj = j0 + 64 * wc + 16 * tb + ((lane & 31) >> 1);
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+#pragma unroll
+                    for (int p = 0; p < 2; ++p) {
+                        int v0 = acc[ta][tb][4 * g + 2 * p];
+                        int v1 = acc[ta][tb][4 * g + 2 * p + 1];
+                        int sv0 = __shfl_xor(v0, 1);
+                        int sv1 = __shfl_xor(v1, 1);
+                        long i = ci_base + 4 * g + 2 * (lane >> 5) + p;
+                        long j = cj;
+                        bool write = (lane & 1) == 0 && i < n && j < n &&
+                                     i >= j;
+                        if (write) {
+                            float re = (float)(v0 + sv1);
+                            float im = (float)(sv0 - v1);
+                            f2 prev =
+                                beta != 0.f ? cb[i * c_row + j] : f2{};
+                            cb[i * c_row + j] =
+                                f2{alpha * re + beta * prev.x,
+                                   alpha * im + beta * prev.y};
+                        }
+                    }
+                }
+            }
+        }
+    }
+}
+
 /* -------- 8-wave rectangular-tile register-staged cherk (rs8) ----------- */
 // 512 threads / 8 waves per 128x64-complex output tile (wave grid 4x2,
 // each wave the same 32x32-complex quadrant as the rs kernel).  Staged
@@ -1913,14 +2133,38 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                         ((a_k * 2) % 16 == 0) && ((a_b * 2) % 16 == 0);
             bool want_rs = !sel || strcmp(sel, "rs") == 0;
             bool want_wave = sel && strcmp(sel, "wave") == 0;
-            // rs2 big-tile kernel: default when eligible (round-2 winner);
-            // BIFROST_CHERK=rs forces the round-1 kernel.
-            bool want_rs2 = (!sel || strcmp(sel, "rs2") == 0) &&
+            // rs3 (32x32x32-MFMA big tile): default when eligible;
+            // BIFROST_CHERK=rs2 selects the 16x16x64 big-tile form,
+            // =rs the round-1 kernel.
+            bool want_rs3 = (!sel || strcmp(sel, "rs3") == 0) &&
+                            n % 128 == 0;
+            bool want_rs2 = sel && strcmp(sel, "rs2") == 0 &&
                             n % 128 == 0;
             // rs8 (8-wave 128x64 tile) measures ~equal to rs (1.23 vs
             // 1.25 Gsamp/s at config 3) — opt-in until it wins.
             const char* sel8 = getenv("BIFROST_CHERK");
             bool want_rs8 = sel8 && strcmp(sel8, "rs8") == 0;
+            if (al16 && want_rs3) {
+                long nb3 = n / 128;
+                long ntiles3 = nb3 * (nb3 + 1) / 2;
+                long nflat3 = ((ntiles3 * nbatch + 7) / 8) * 8;
+                dim3 grid3(cap_grid(nflat3, 65535), 1);
+                const char* schenv3 = getenv("BIFROST_CHERK_SCHED");
+                int sched3 = schenv3 ? atoi(schenv3) : 1;
+                auto launch_rs3 = [&](auto kern) {
+                    hipLaunchKernelGGL(kern, grid3, dim3(256), 0, stream, n,
+                                       k, nbatch, (float)alpha,
+                                       (const signed char*)a, a_k, a_b,
+                                       (float)beta, (f2*)c, c_row, c_b,
+                                       ntiles3);
+                };
+                if (sched3 == 0)
+                    launch_rs3(cherk_ci8_mfma32_rs3_kernel<0>);
+                else
+                    launch_rs3(cherk_ci8_mfma32_rs3_kernel<1>);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
             if (al16 && want_rs2) {
                 long nb2 = n / 128;
                 long ntiles2 = nb2 * (nb2 + 1) / 2;
