@@ -2151,6 +2151,8 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 dim3 grid3(cap_grid(nflat3, 65535), 1);
                 const char* schenv3 = getenv("BIFROST_CHERK_SCHED");
                 int sched3 = schenv3 ? atoi(schenv3) : 1;
+                const char* genv = getenv("BIFROST_CHERK_GRID");
+                if (genv) grid3 = dim3(cap_grid(atol(genv), 65535), 1);
                 auto launch_rs3 = [&](auto kern) {
                     hipLaunchKernelGGL(kern, grid3, dim3(256), 0, stream, n,
                                        k, nbatch, (float)alpha,
@@ -2172,6 +2174,8 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 dim3 grid2(cap_grid(nflat2, 65535), 1);
                 const char* schenv = getenv("BIFROST_CHERK_SCHED");
                 int sched = schenv ? atoi(schenv) : 0;
+                const char* genv2 = getenv("BIFROST_CHERK_GRID");
+                if (genv2) grid2 = dim3(cap_grid(atol(genv2), 65535), 1);
                 auto launch_rs2 = [&](auto kern) {
                     hipLaunchKernelGGL(kern, grid2, dim3(256), 0, stream, n,
                                        k, nbatch, (float)alpha,

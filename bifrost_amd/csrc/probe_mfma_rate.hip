@@ -89,6 +89,57 @@ void mfma_pairs_kernel(const int* __restrict__ in, long* __restrict__ cyc,
     }
 }
 
+// Kernel-convergence variants: reproduce the in-situ slowdown by adding
+// the rs2 kernel's environment piece by piece.
+//   LDSALLOC: declare the kernel's 69632-B LDS (occupancy via LDS cap)
+//   OUTER:    20-tile outer loop with per-"slab" __syncthreads
+template <int LDSALLOC, int OUTER, int WPE>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(WPE)))
+void mfma_pairs_env_kernel(const int* __restrict__ in,
+                           long* __restrict__ cyc, int* __restrict__ sink,
+                           int iter) {
+    extern __shared__ signed char dyn_lds[];
+    v4i acc[8][8];
+    v4i fa[8], fb[8];
+    int lane = threadIdx.x & 63;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+        fa[u] = v4i{in[lane + u], in[lane + u + 64], in[lane + u + 128],
+                    in[lane + u + 192]};
+        fb[u] = v4i{in[lane + u + 1], in[lane + u + 65],
+                    in[lane + u + 129], in[lane + u + 193]};
+    }
+#pragma unroll
+    for (int x = 0; x < 8; ++x)
+#pragma unroll
+        for (int y = 0; y < 8; ++y) acc[x][y] = v4i{in[x], 0, 0, in[y]};
+    if (LDSALLOC && threadIdx.x == 0) dyn_lds[0] = (signed char)in[0];
+    __syncthreads();
+    long t0 = __builtin_amdgcn_s_memtime();
+    for (int tile = 0; tile < (OUTER ? 20 : 1); ++tile) {
+        for (int it = 0; it < iter; ++it) {
+#pragma unroll
+            for (int ta = 0; ta < 8; ++ta)
+#pragma unroll
+                for (int tb = 0; tb < 8; ++tb)
+                    acc[ta][tb] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+                        fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+            if (OUTER) __syncthreads();
+        }
+    }
+    long t1 = __builtin_amdgcn_s_memtime();
+    int s = 0;
+#pragma unroll
+    for (int x = 0; x < 8; ++x)
+#pragma unroll
+        for (int y = 0; y < 8; ++y) s += acc[x][y][0] + acc[x][y][3];
+    if (threadIdx.x == 0) {
+        sink[blockIdx.x] = s;
+        cyc[blockIdx.x] = (t1 - t0) / (OUTER ? 20 : 1);
+    }
+}
+
 // rsm geometry: fa[8] x fb[4] -> 32 accumulators (128 regs) at
 // 2 waves/SIMD — do co-resident waves hide the operand-switch stalls?
 template <int WPE>
@@ -232,7 +283,8 @@ void mfma_tr8_kernel(const int* __restrict__ in, long* __restrict__ cyc,
 }
 
 template <typename K>
-void run(const char* name, K kern, int blocks, int iter, int nacc) {
+void run(const char* name, K kern, int blocks, int iter, int nacc,
+         int ldsbytes = 0) {
     int* in;
     long* cyc;
     int* sink;
@@ -240,11 +292,11 @@ void run(const char* name, K kern, int blocks, int iter, int nacc) {
     (void)hipMalloc(&cyc, blocks * sizeof(long));
     (void)hipMalloc(&sink, blocks * sizeof(int));
     (void)hipMemset(in, 1, 4096);
-    hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, 0, in, cyc, sink,
-                       iter);
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), ldsbytes, 0, in, cyc,
+                       sink, iter);
     (void)hipDeviceSynchronize();
-    hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, 0, in, cyc, sink,
-                       iter);
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), ldsbytes, 0, in, cyc,
+                       sink, iter);
     (void)hipDeviceSynchronize();
     long* h = new long[blocks];
     (void)hipMemcpy(h, cyc, blocks * sizeof(long), hipMemcpyDeviceToHost);
@@ -271,6 +323,13 @@ int main() {
     run("occ4 nacc8", mfma_rate_kernel<8, 4>, 1024, iter, 8);
     // diverse 8x8 operand pairs (the rs2 burst register pattern)
     run("pairs occ1 (rs2 burst)", mfma_pairs_kernel<1>, 256, iter, 64);
+    run("env base (pairs occ1)", mfma_pairs_env_kernel<0, 0, 1>, 256,
+        iter, 64);
+    run("env +lds69k", mfma_pairs_env_kernel<1, 0, 1>, 256, iter, 64,
+        69632);
+    run("env +outer/sync", mfma_pairs_env_kernel<0, 1, 1>, 256, 500, 64);
+    run("env +lds+outer", mfma_pairs_env_kernel<1, 1, 1>, 256, 500, 64,
+        69632);
     run("pairs2 occ2 (rsm 8x4)", mfma_pairs2_kernel<2>, 512, iter, 32);
     run("pairs2 occ1 (8x4)", mfma_pairs2_kernel<1>, 256, iter, 32);
     // 32x32x32 i8 (16 i32 acc per instr; cyc shown PER 16x16x64-EQUIV,

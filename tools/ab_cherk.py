@@ -30,10 +30,13 @@ WARM = 3
 def main():
     # a leading "!" marks a DIAGNOSTIC variant (wrong results by design,
     # e.g. compute-only/staging-only ablations): parity check is skipped.
+    # spec = NAME:CHERK:SCHED[:GRID]
     variants = []
     for spec in sys.argv[1:]:
-        name, cherk, sched = spec.split(":")
-        variants.append((name, cherk or None, sched or None))
+        parts = spec.split(":")
+        name, cherk, sched = parts[0], parts[1], parts[2]
+        grid = parts[3] if len(parts) > 3 else None
+        variants.append((name, cherk or None, sched or None, grid))
     if not variants:
         variants = [
             ("rs2s0", "rs2", "0"),
@@ -67,17 +70,20 @@ def main():
     tri = np.triu_indices(N, 1)
     gold2[..., tri[0], tri[1]] = 0
 
-    def set_env(cherk, sched):
-        for k in ("BIFROST_CHERK", "BIFROST_CHERK_SCHED"):
+    def set_env(cherk, sched, grid=None):
+        for k in ("BIFROST_CHERK", "BIFROST_CHERK_SCHED",
+                  "BIFROST_CHERK_GRID"):
             os.environ.pop(k, None)
         if cherk:
             os.environ["BIFROST_CHERK"] = cherk
         if sched:
             os.environ["BIFROST_CHERK_SCHED"] = sched
+        if grid:
+            os.environ["BIFROST_CHERK_GRID"] = grid
 
     results = {}
-    for name, cherk, sched in variants:
-        set_env(cherk, sched)
+    for name, cherk, sched, grid in variants:
+        set_env(cherk, sched, grid)
         if name.startswith("!"):
             linalg.matmul(1, None, x_view, 0, vis)  # warm compile path
             torch.cuda.synchronize()
@@ -95,10 +101,10 @@ def main():
                               "max_abs_err": float(bad.max())}))
 
     for rep in range(REPS):
-        for name, cherk, sched in variants:
+        for name, cherk, sched, grid in variants:
             if not results[name]["parity"]:
                 continue
-            set_env(cherk, sched)
+            set_env(cherk, sched, grid)
             for _ in range(WARM):
                 linalg.matmul(1, None, x_view, 1, vis)
             torch.cuda.synchronize()
