@@ -1382,6 +1382,223 @@ void cherk_ci8_mfma32_rs3_kernel(long n, long k, long nbatch, float alpha,
     }
 }
 
+/* ---- rs4: occupancy-2 mid-tile cherk on v_mfma_i32_32x32x32_i8 -------- */
+// The occ-1 big-tile kernels (rs2/rs3) sit ~3x above the probed MFMA
+// floor: the same diverse-operand MFMA stream measures 16-28 cyc/instr-
+// equiv in isolated probes but ~49 in situ at 1 wave/SIMD — an issue-
+// side stall that nothing can hide at occupancy 1.  The same probe shows
+// a CO-RESIDENT SECOND WAVE hides it completely (pairs at occ2: 16.4
+// cyc/MFMA per SIMD, the full pipe rate).  rs4 therefore halves the
+// per-wave tile to 64x32 complex (4x2 grid of 32x32-byte mfma32 tiles,
+// 8 x v16i = 128 accumulator regs) so TWO workgroups co-reside
+// (2 waves/SIMD, LDS 2x53 KB); the partner workgroup's MFMAs fill every
+// stall.  Strips: I [64][272] as rs3; J [64][144] (128 B + 16-B pad,
+// conflict-free by the same bank argument).  Workgroup tile 128x64
+// complex over a rectangular-triangular tile map (row blocks of 128,
+// col blocks of 64, J <= 2I+1 — the rs8 map).  Fragment gather and
+// epilogue layout identical to rs3 (probe-verified).
+#define RS4_JROW 144
+#define RS4_ISTRIP (64 * RS2_ROW)
+#define RS4_JSTRIP (64 * RS4_JROW)
+template <int SCHED>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(2)))
+void cherk_ci8_mfma32_rs4_kernel(long n, long k, long nbatch, float alpha,
+                                 const signed char* __restrict__ a, long lda,
+                                 long a_b, float beta, f2* __restrict__ c,
+                                 long c_row, long c_b, long nti,
+                                 long ntiles) {
+    __shared__ signed char ldsI[2][64][RS2_ROW];
+    __shared__ signed char ldsJ[2][64][RS4_JROW];
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;  // row half (64c) x col half (32c)
+
+    int row32 = 16 * (lane >> 5) + ((lane >> 1) & 7);
+    int colb32 = 8 * ((lane & 1) + 2 * ((lane >> 4) & 1));
+
+    // staging: threads 0-127 stage the I strip (128 B each, chunks
+    // 2e+h), threads 128-255 the J strip (64 B each, chunks 2e+h)
+    int st_isJ = tid >> 7;
+    int tt = tid & 127;
+    int st_row = tt >> 1;
+    int st_h = tt & 1;
+
+    const signed char* rdI0 = &ldsI[0][0][0] + row32 * RS2_ROW + colb32
+                            + 128 * wr;
+    const signed char* rdJ0 = &ldsJ[0][0][0] + row32 * RS4_JROW + colb32
+                            + 64 * wc;
+    signed char* wrI0 = &ldsI[0][0][0] + st_row * RS2_ROW + 16 * st_h;
+    signed char* wrJ0 = &ldsJ[0][0][0] + st_row * RS4_JROW + 16 * st_h;
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        // rectangular lower-triangle map: row block I (128 complex),
+        // col block J (64 complex), J <= min(2I+1, nti-1)
+        long I = 0, rem = t;
+        while (rem >= (2 * I + 2 < nti ? 2 * I + 2 : nti)) {
+            rem -= (2 * I + 2 < nti ? 2 * I + 2 : nti);
+            ++I;
+        }
+        long J = rem;
+        long i0 = I * 128, j0 = J * 64;
+        // quadrant diagonal classification (d multiple of 32):
+        //   d >= 64: all above -> skip; d <= -32: full; else checked
+        long d = (j0 + 32 * wc) - (i0 + 64 * wr);
+        bool skip_all = d >= 64;
+        bool crossing = !skip_all && d > -32;
+        int dq = (int)d;
+        v16i acc[4][2];
+#pragma unroll
+        for (int x = 0; x < 4; ++x)
+#pragma unroll
+            for (int y = 0; y < 2; ++y) acc[x][y] = v16i{};
+
+        const long slab_step = 64 * lda * 2;
+        v4i stg[8];
+        const signed char* load_next = st_isJ
+            ? ab + (long)st_row * lda * 2 + j0 * 2 + 16 * st_h
+            : ab + (long)st_row * lda * 2 + i0 * 2 + 16 * st_h;
+        auto load_slab = [&]() {
+            const signed char* p = load_next;
+            load_next += slab_step;
+            if (st_isJ) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e)
+                    stg[e] = *(const v4i*)__builtin_assume_aligned(
+                        p + 32 * e, 16);
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    stg[e] = *(const v4i*)__builtin_assume_aligned(
+                        p + 32 * e, 16);
+            }
+        };
+        auto write_slab = [&](int buf) {
+            if (st_isJ) {
+                signed char* wb = wrJ0 + buf * RS4_JSTRIP;
+#pragma unroll
+                for (int e = 0; e < 4; ++e)
+                    *(v4i*)(wb + 16 * (2 * e)) = stg[e];
+            } else {
+                signed char* wb = wrI0 + buf * RS4_ISTRIP;
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    *(v4i*)(wb + 16 * (2 * e)) = stg[e];
+            }
+        };
+        auto fragI = [&](const signed char* base, int h, int cc) {
+            const signed char* p = base + (32 * h) * RS2_ROW + 32 * cc;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 8 * RS2_ROW));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto fragJ = [&](const signed char* base, int h, int cc) {
+            const signed char* p = base + (32 * h) * RS4_JROW + 32 * cc;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 8 * RS4_JROW));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        // per-k-half burst: 6 fragment reads feed 8 mfma32
+        auto burst = [&](const signed char* bI, const signed char* bJ,
+                         bool checked) {
+#pragma unroll
+            for (int h = 0; h < 2; ++h) {
+                v4i fb[2], fa[4];
+#pragma unroll
+                for (int u = 0; u < 2; ++u) fb[u] = fragJ(bJ, h, u);
+#pragma unroll
+                for (int u = 0; u < 4; ++u) fa[u] = fragI(bI, h, u);
+#pragma unroll
+                for (int ta = 0; ta < 4; ++ta)
+#pragma unroll
+                    for (int tb = 0; tb < 2; ++tb) {
+                        if (checked && 16 * ta + 15 < 16 * tb + dq)
+                            continue;
+                        acc[ta][tb] = __builtin_amdgcn_mfma_i32_32x32x32_i8(
+                            fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+                    }
+            }
+        };
+        auto step = [&](int buf, int s, int nslab) {
+            const signed char* bI = &ldsI[buf][0][0] + (rdI0 -
+                                    &ldsI[0][0][0]);
+            const signed char* bJ = &ldsJ[buf][0][0] + (rdJ0 -
+                                    &ldsJ[0][0][0]);
+            if (SCHED == 0) {
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+                if (!skip_all) burst(bI, bJ, crossing);
+            } else {
+                if (!skip_all) burst(bI, bJ, crossing);
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+            }
+            __syncthreads();
+        };
+
+        int nslab = (int)(k / 64);
+        load_slab();
+        write_slab(0);
+        if (nslab > 1) load_slab();
+        __syncthreads();
+        int s = 0;
+        while (s < nslab) {
+            step(0, s, nslab);
+            ++s;
+            if (s >= nslab) break;
+            step(1, s, nslab);
+            ++s;
+        }
+#pragma unroll
+        for (int ta = 0; ta < 4; ++ta) {
+#pragma unroll
+            for (int tb = 0; tb < 2; ++tb) {
+                if (skip_all || (crossing && 16 * ta + 15 < 16 * tb + dq))
+                    continue;
+                long ci_base = i0 + 64 * wr + 16 * ta;
+                long cj = j0 + 32 * wc + 16 * tb + ((lane & 31) >> 1);
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+#pragma unroll
+                    for (int p = 0; p < 2; ++p) {
+                        int v0 = acc[ta][tb][4 * g + 2 * p];
+                        int v1 = acc[ta][tb][4 * g + 2 * p + 1];
+                        int sv0 = __shfl_xor(v0, 1);
+                        int sv1 = __shfl_xor(v1, 1);
+                        long i = ci_base + 4 * g + 2 * (lane >> 5) + p;
+                        long j = cj;
+                        bool write = (lane & 1) == 0 && i < n && j < n &&
+                                     i >= j;
+                        if (write) {
+                            float re = (float)(v0 + sv1);
+                            float im = (float)(sv0 - v1);
+                            f2 prev =
+                                beta != 0.f ? cb[i * c_row + j] : f2{};
+                            cb[i * c_row + j] =
+                                f2{alpha * re + beta * prev.x,
+                                   alpha * im + beta * prev.y};
+                        }
+                    }
+                }
+            }
+        }
+    }
+}
+
 /* -------- 8-wave rectangular-tile register-staged cherk (rs8) ----------- */
 // 512 threads / 8 waves per 128x64-complex output tile (wave grid 4x2,
 // each wave the same 32x32-complex quadrant as the rs kernel).  Staged
@@ -2138,12 +2355,40 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
             // =rs the round-1 kernel.
             bool want_rs3 = (!sel || strcmp(sel, "rs3") == 0) &&
                             n % 128 == 0;
+            bool want_rs4 = sel && strcmp(sel, "rs4") == 0 &&
+                            n % 128 == 0;
             bool want_rs2 = sel && strcmp(sel, "rs2") == 0 &&
                             n % 128 == 0;
             // rs8 (8-wave 128x64 tile) measures ~equal to rs (1.23 vs
             // 1.25 Gsamp/s at config 3) — opt-in until it wins.
             const char* sel8 = getenv("BIFROST_CHERK");
             bool want_rs8 = sel8 && strcmp(sel8, "rs8") == 0;
+            if (al16 && want_rs4) {
+                long nti4 = n / 64;
+                long nI4 = n / 128;
+                long ntiles4 = 0;
+                for (long I = 0; I < nI4; ++I)
+                    ntiles4 += (2 * I + 2 < nti4 ? 2 * I + 2 : nti4);
+                long nflat4 = ((ntiles4 * nbatch + 7) / 8) * 8;
+                dim3 grid4(cap_grid(nflat4, 65535), 1);
+                const char* genv4 = getenv("BIFROST_CHERK_GRID");
+                if (genv4) grid4 = dim3(cap_grid(atol(genv4), 65535), 1);
+                const char* schenv4 = getenv("BIFROST_CHERK_SCHED");
+                int sched4 = schenv4 ? atoi(schenv4) : 0;
+                auto launch_rs4 = [&](auto kern) {
+                    hipLaunchKernelGGL(kern, grid4, dim3(256), 0, stream, n,
+                                       k, nbatch, (float)alpha,
+                                       (const signed char*)a, a_k, a_b,
+                                       (float)beta, (f2*)c, c_row, c_b,
+                                       nti4, ntiles4);
+                };
+                if (sched4 == 1)
+                    launch_rs4(cherk_ci8_mfma32_rs4_kernel<1>);
+                else
+                    launch_rs4(cherk_ci8_mfma32_rs4_kernel<0>);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
             if (al16 && want_rs3) {
                 long nb3 = n / 128;
                 long ntiles3 = nb3 * (nb3 + 1) / 2;
