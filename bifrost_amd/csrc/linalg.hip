@@ -1043,7 +1043,7 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
         // the k-loop's MFMA stream runs with the real register pattern
         // but zero LDS reads.
         v4i ffa[8], ffb[8];
-        if (SCHED == 9 || SCHED == 11) {
+        if (SCHED == 9 || SCHED == 11 || SCHED == 13) {
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
                 ffb[u] = frag(rdJ0, u);
@@ -1070,6 +1070,10 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
                 } else {
                     burst_full(bI, bJ);
                 }
+                return;
+            } else if (SCHED == 13) {
+                // DIAGNOSTIC: pure fixed-frag stream + k-loop s_memtime
+                if (!skip_all) burst_fixed();
                 return;
             } else if (SCHED == 9) {
                 // DIAGNOSTIC: fixed-fragment MFMA stream + barrier
@@ -1106,7 +1110,7 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
                 } else {
                     burst_full(bI, bJ);
                 }
-            } else {
+            } else {  // 1 and 12: burst first, then writes+loads
                 if (skip_all) {
                 } else if (diag_q) {
                     burst_diag(bI, bJ);
@@ -1126,6 +1130,8 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
         write_slab(wr0);
         if (nslab > 1) load_slab();
         __syncthreads();
+        long tt0 = 0;
+        if (SCHED >= 12) tt0 = __builtin_amdgcn_s_memtime();
         int s = 0;
         while (s < nslab) {
             step(0, s, nslab);
@@ -1133,6 +1139,11 @@ void cherk_ci8_mfma_rs2_kernel(long n, long k, long nbatch, float alpha,
             if (s >= nslab) break;
             step(1, s, nslab);
             ++s;
+        }
+        if (SCHED >= 12) {
+            long tt1 = __builtin_amdgcn_s_memtime();
+            if (lane == 0)
+                ((long long*)c)[(blockIdx.x & 16383) * 4 + wave] = tt1 - tt0;
         }
 #pragma unroll
         for (int ta = 0; ta < 8; ++ta) {
@@ -1570,6 +1581,178 @@ void cherk_ci8_mfma32_rs4_kernel(long n, long k, long nbatch, float alpha,
                 if (skip_all || (crossing && 16 * ta + 15 < 16 * tb + dq))
                     continue;
                 long ci_base = i0 + 64 * wr + 16 * ta;
+                long cj = j0 + 32 * wc + 16 * tb + ((lane & 31) >> 1);
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+#pragma unroll
+                    for (int p = 0; p < 2; ++p) {
+                        int v0 = acc[ta][tb][4 * g + 2 * p];
+                        int v1 = acc[ta][tb][4 * g + 2 * p + 1];
+                        int sv0 = __shfl_xor(v0, 1);
+                        int sv1 = __shfl_xor(v1, 1);
+                        long i = ci_base + 4 * g + 2 * (lane >> 5) + p;
+                        long j = cj;
+                        bool write = (lane & 1) == 0 && i < n && j < n &&
+                                     i >= j;
+                        if (write) {
+                            float re = (float)(v0 + sv1);
+                            float im = (float)(sv0 - v1);
+                            f2 prev =
+                                beta != 0.f ? cb[i * c_row + j] : f2{};
+                            cb[i * c_row + j] =
+                                f2{alpha * re + beta * prev.x,
+                                   alpha * im + beta * prev.y};
+                        }
+                    }
+                }
+            }
+        }
+    }
+}
+
+/* ---- rs5: occupancy-3 small-tile cherk on v_mfma_i32_32x32x32_i8 ------ */
+// Occupancy ladder endpoint: rs2/rs3 (occ1) exposed an in-situ MFMA
+// issue stall; rs4 (occ2) recovered to ~1.1 Gsamp/s; round-1 rs at
+// occ 3-4 still led — occupancy dominates every tile-shape refinement.
+// rs5 takes the rs geometry (64x64-complex workgroup tile, 32x32 per
+// wave, triangular 64-block map) but runs it on 32x32x32 i8 MFMAs with
+// the round-2 zero-VALU addressing: 8 instructions per wave-slab
+// (vs rs's 16), 64 accumulator regs, ~150-reg footprint -> 3 waves/SIMD
+// with 4-WG LDS headroom (both strips [64][144]).
+#define RS5_ROW 144
+#define RS5_STRIP (64 * RS5_ROW)
+#define RS5_BUF (2 * RS5_STRIP)
+template <int SCHED>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(3)))
+void cherk_ci8_mfma32_rs5_kernel(long n, long k, long nbatch, float alpha,
+                                 const signed char* __restrict__ a, long lda,
+                                 long a_b, float beta, f2* __restrict__ c,
+                                 long c_row, long c_b, long ntiles) {
+    __shared__ signed char lds[2][2][64][RS5_ROW];  // [buf][strip][k][b]
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;  // 32-complex quadrant row/col
+
+    int row32 = 16 * (lane >> 5) + ((lane >> 1) & 7);
+    int colb32 = 8 * ((lane & 1) + 2 * ((lane >> 4) & 1));
+
+    // staging: 128 threads per strip, 64 B per thread (chunks 2e+h)
+    int st_strip = tid >> 7;
+    int tt = tid & 127;
+    int st_row = tt >> 1;
+    int st_h = tt & 1;
+
+    signed char* lds0 = &lds[0][0][0][0];
+    const signed char* rdI0 = lds0 + row32 * RS5_ROW + colb32 + 64 * wr;
+    const signed char* rdJ0 = lds0 + RS5_STRIP + row32 * RS5_ROW + colb32
+                            + 64 * wc;
+    signed char* wr0 = lds0 + st_strip * RS5_STRIP + st_row * RS5_ROW
+                     + 16 * st_h;
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        long bi, bj;
+        lift_tri(t, bi, bj);
+        long i0 = bi * 64, j0 = bj * 64;
+        bool diag = bi == bj;
+        bool skip_all = diag && wr < wc;
+        bool diag_q = diag && wr == wc;  // skip the (ta=0,tb=1) tile
+        v16i acc[2][2];
+#pragma unroll
+        for (int x = 0; x < 2; ++x)
+#pragma unroll
+            for (int y = 0; y < 2; ++y) acc[x][y] = v16i{};
+
+        long base_col = st_strip ? j0 : i0;
+        const long slab_step = 64 * lda * 2;
+        v4i stg[4];
+        const signed char* load_next = ab + (long)st_row * lda * 2 +
+                                       base_col * 2 + 16 * st_h;
+        auto load_slab = [&]() {
+            const signed char* p = load_next;
+            load_next += slab_step;
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+                stg[e] = *(const v4i*)__builtin_assume_aligned(p + 32 * e,
+                                                               16);
+        };
+        auto write_slab = [&](int buf) {
+            signed char* wb = wr0 + buf * RS5_BUF;
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+                *(v4i*)(wb + 16 * (2 * e)) = stg[e];
+        };
+        auto frag = [&](const signed char* base, int h, int cc) {
+            const signed char* p = base + (32 * h) * RS5_ROW + 32 * cc;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 8 * RS5_ROW));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto burst = [&](const signed char* bI, const signed char* bJ) {
+#pragma unroll
+            for (int h = 0; h < 2; ++h) {
+                v4i fb[2], fa[2];
+                fb[0] = frag(bJ, h, 0);
+                fb[1] = frag(bJ, h, 1);
+                fa[0] = frag(bI, h, 0);
+                fa[1] = frag(bI, h, 1);
+#pragma unroll
+                for (int ta = 0; ta < 2; ++ta)
+#pragma unroll
+                    for (int tb = 0; tb < 2; ++tb) {
+                        if (diag_q && ta < tb) continue;
+                        acc[ta][tb] = __builtin_amdgcn_mfma_i32_32x32x32_i8(
+                            fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+                    }
+            }
+        };
+        auto step = [&](int buf, int s, int nslab) {
+            const signed char* bI = rdI0 + buf * RS5_BUF;
+            const signed char* bJ = rdJ0 + buf * RS5_BUF;
+            if (SCHED == 0) {
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+                if (!skip_all) burst(bI, bJ);
+            } else {
+                if (!skip_all) burst(bI, bJ);
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+            }
+            __syncthreads();
+        };
+
+        int nslab = (int)(k / 64);
+        load_slab();
+        write_slab(0);
+        if (nslab > 1) load_slab();
+        __syncthreads();
+        int s = 0;
+        while (s < nslab) {
+            step(0, s, nslab);
+            ++s;
+            if (s >= nslab) break;
+            step(1, s, nslab);
+            ++s;
+        }
+#pragma unroll
+        for (int ta = 0; ta < 2; ++ta) {
+#pragma unroll
+            for (int tb = 0; tb < 2; ++tb) {
+                if (skip_all || (diag_q && ta < tb)) continue;
+                long ci_base = i0 + 32 * wr + 16 * ta;
                 long cj = j0 + 32 * wc + 16 * tb + ((lane & 31) >> 1);
 #pragma unroll
                 for (int g = 0; g < 4; ++g) {
@@ -2357,12 +2540,31 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                             n % 128 == 0;
             bool want_rs4 = sel && strcmp(sel, "rs4") == 0 &&
                             n % 128 == 0;
+            bool want_rs5 = sel && strcmp(sel, "rs5") == 0;
             bool want_rs2 = sel && strcmp(sel, "rs2") == 0 &&
                             n % 128 == 0;
             // rs8 (8-wave 128x64 tile) measures ~equal to rs (1.23 vs
             // 1.25 Gsamp/s at config 3) — opt-in until it wins.
             const char* sel8 = getenv("BIFROST_CHERK");
             bool want_rs8 = sel8 && strcmp(sel8, "rs8") == 0;
+            if (al16 && want_rs5) {
+                const char* schenv5 = getenv("BIFROST_CHERK_SCHED");
+                int sched5 = schenv5 ? atoi(schenv5) : 0;
+                if (sched5 == 1)
+                    hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<1>, grid,
+                                       dim3(256), 0, stream, n, k, nbatch,
+                                       (float)alpha, (const signed char*)a,
+                                       a_k, a_b, (float)beta, (f2*)c, c_row,
+                                       c_b, ntiles);
+                else
+                    hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<0>, grid,
+                                       dim3(256), 0, stream, n, k, nbatch,
+                                       (float)alpha, (const signed char*)a,
+                                       a_k, a_b, (float)beta, (f2*)c, c_row,
+                                       c_b, ntiles);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
             if (al16 && want_rs4) {
                 long nti4 = n / 64;
                 long nI4 = n / 128;
@@ -2440,6 +2642,10 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<9>);
                 else if (sched == 11)  // diagnostic: wrong results
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<11>);
+                else if (sched == 12)  // diagnostic: k-loop cycle dump
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<12>);
+                else if (sched == 13)  // diagnostic: pure + cycle dump
+                    launch_rs2(cherk_ci8_mfma_rs2_kernel<13>);
                 else
                     launch_rs2(cherk_ci8_mfma_rs2_kernel<0>);
                 BF_CHECK_HIP(hipGetLastError());

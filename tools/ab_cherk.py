@@ -87,6 +87,8 @@ def main():
         if name.startswith("!"):
             linalg.matmul(1, None, x_view, 0, vis)  # warm compile path
             torch.cuda.synchronize()
+            if sched in ("12", "13"):
+                dump_cycles(vis, 5120, name)
             results[name] = {"parity": True, "gsps": []}
             continue
         linalg.matmul(1, None, x_view, 0, vis)
@@ -128,3 +130,19 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def dump_cycles(vis, nblocks, label):
+    """Read per-wave k-loop shader cycles left in the vis buffer by the
+    SCHED>=12 diagnostic kernels (first 4*nblocks int64 slots)."""
+    import numpy as np
+    raw = np.asarray(vis.copy("system")).view(np.int64).ravel()[:4 * nblocks]
+    raw = raw[raw > 0]
+    if not len(raw):
+        print(label, "no cycle records")
+        return
+    print(label, "k-loop shader cycles/wave: median %d  p10 %d  p90 %d  "
+          "(per-slab %.0f, per-instr-equiv %.1f over 64 slabs x 64 eq)"
+          % (int(np.median(raw)), int(np.percentile(raw, 10)),
+             int(np.percentile(raw, 90)), np.median(raw) / 64.0,
+             np.median(raw) / 64.0 / 64.0))
