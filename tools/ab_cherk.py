@@ -27,6 +27,22 @@ STEPS = 10
 WARM = 3
 
 
+def dump_cycles(vis, nblocks, label):
+    """Read per-wave k-loop shader cycles left in the vis buffer by the
+    SCHED>=12 diagnostic kernels (first 4*nblocks int64 slots)."""
+    import numpy as np
+    raw = np.asarray(vis.copy("system")).view(np.int64).ravel()[:4 * nblocks]
+    raw = raw[raw > 0]
+    if not len(raw):
+        print(label, "no cycle records")
+        return
+    print(label, "k-loop shader cycles/wave: median %d  p10 %d  p90 %d  "
+          "(per-slab %.0f, per-instr-equiv %.1f over 64 slabs x 64 eq)"
+          % (int(np.median(raw)), int(np.percentile(raw, 10)),
+             int(np.percentile(raw, 90)), np.median(raw) / 64.0,
+             np.median(raw) / 64.0 / 64.0))
+
+
 def main():
     # a leading "!" marks a DIAGNOSTIC variant (wrong results by design,
     # e.g. compute-only/staging-only ablations): parity check is skipped.
@@ -131,18 +147,3 @@ def main():
 if __name__ == "__main__":
     main()
 
-
-def dump_cycles(vis, nblocks, label):
-    """Read per-wave k-loop shader cycles left in the vis buffer by the
-    SCHED>=12 diagnostic kernels (first 4*nblocks int64 slots)."""
-    import numpy as np
-    raw = np.asarray(vis.copy("system")).view(np.int64).ravel()[:4 * nblocks]
-    raw = raw[raw > 0]
-    if not len(raw):
-        print(label, "no cycle records")
-        return
-    print(label, "k-loop shader cycles/wave: median %d  p10 %d  p90 %d  "
-          "(per-slab %.0f, per-instr-equiv %.1f over 64 slabs x 64 eq)"
-          % (int(np.median(raw)), int(np.percentile(raw, 10)),
-             int(np.percentile(raw, 90)), np.median(raw) / 64.0,
-             np.median(raw) / 64.0 / 64.0))
