@@ -2533,10 +2533,12 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                         ((a_k * 2) % 16 == 0) && ((a_b * 2) % 16 == 0);
             bool want_rs = !sel || strcmp(sel, "rs") == 0;
             bool want_wave = sel && strcmp(sel, "wave") == 0;
-            // rs3 (32x32x32-MFMA big tile): default when eligible;
-            // BIFROST_CHERK=rs2 selects the 16x16x64 big-tile form,
-            // =rs the round-1 kernel.
-            bool want_rs3 = (!sel || strcmp(sel, "rs3") == 0) &&
+            // Round-2 A/B outcome (profiles/round2_cherk.md): the chip
+            // is POWER-bound on random data; the round-1 rs kernel and
+            // the occ-3/4 mfma32 rs5 tie at the power envelope, the
+            // occ-1 big-tile forms (rs2/rs3) lose to an issue stall
+            // nothing can hide.  Default = rs (round-1), rs5 via env.
+            bool want_rs3 = sel && strcmp(sel, "rs3") == 0 &&
                             n % 128 == 0;
             bool want_rs4 = sel && strcmp(sel, "rs4") == 0 &&
                             n % 128 == 0;

@@ -47,12 +47,15 @@ def main():
     # a leading "!" marks a DIAGNOSTIC variant (wrong results by design,
     # e.g. compute-only/staging-only ablations): parity check is skipped.
     # spec = NAME:CHERK:SCHED[:GRID]
+    # 5th field "Z" = run on ZEROED voltages (same kernel, zero-toggle
+    # operand data) — isolates the DVFS/power effect on sustained clock.
     variants = []
     for spec in sys.argv[1:]:
         parts = spec.split(":")
         name, cherk, sched = parts[0], parts[1], parts[2]
-        grid = parts[3] if len(parts) > 3 else None
-        variants.append((name, cherk or None, sched or None, grid))
+        grid = parts[3] if len(parts) > 3 and parts[3] else None
+        zero = len(parts) > 4 and parts[4] == "Z"
+        variants.append((name, cherk or None, sched or None, grid, zero))
     if not variants:
         variants = [
             ("rs2s0", "rs2", "0"),
@@ -77,6 +80,9 @@ def main():
                                   .reshape(NTIME, NCHAN, N)), space="cuda")
     x_view = x_dev.transpose(1, 0, 2)
     vis = bf.ndarray(space="cuda", shape=(NCHAN, N, N), dtype="cf32")
+    x0_dev = bf.asarray(bf.ndarray(np.zeros_like(x8).view(bf.DataType.ci8)
+                                   .reshape(NTIME, NCHAN, N)), space="cuda")
+    x0_view = x0_dev.transpose(1, 0, 2)
     linalg = LinAlg()
 
     # small-slice parity gold (first 2 channels, full k)
@@ -98,8 +104,11 @@ def main():
             os.environ["BIFROST_CHERK_GRID"] = grid
 
     results = {}
-    for name, cherk, sched, grid in variants:
+    for name, cherk, sched, grid, zero in variants:
         set_env(cherk, sched, grid)
+        if zero:
+            results[name] = {"parity": True, "gsps": []}
+            continue
         if name.startswith("!"):
             linalg.matmul(1, None, x_view, 0, vis)  # warm compile path
             torch.cuda.synchronize()
@@ -119,16 +128,17 @@ def main():
                               "max_abs_err": float(bad.max())}))
 
     for rep in range(REPS):
-        for name, cherk, sched, grid in variants:
+        for name, cherk, sched, grid, zero in variants:
             if not results[name]["parity"]:
                 continue
             set_env(cherk, sched, grid)
+            xv = x0_view if zero else x_view
             for _ in range(WARM):
-                linalg.matmul(1, None, x_view, 1, vis)
+                linalg.matmul(1, None, xv, 1, vis)
             torch.cuda.synchronize()
             t0 = time.perf_counter()
             for _ in range(STEPS):
-                linalg.matmul(1, None, x_view, 1, vis)
+                linalg.matmul(1, None, xv, 1, vis)
             torch.cuda.synchronize()
             dt = time.perf_counter() - t0
             gsps = STEPS * NTIME * NCHAN / dt / 1e9
