@@ -2542,7 +2542,12 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                             n % 128 == 0;
             bool want_rs4 = sel && strcmp(sel, "rs4") == 0 &&
                             n % 128 == 0;
-            bool want_rs5 = sel && strcmp(sel, "rs5") == 0;
+            // rs5 is the round-2 default: ties rs on power-bound
+            // random data, +3% on the zero-toggle ceiling, half the
+            // instruction stream (profiles/round2_cherk.md); rs remains
+            // the fallback for shapes rs5 cannot take and via
+            // BIFROST_CHERK=rs.
+            bool want_rs5 = !sel || strcmp(sel, "rs5") == 0;
             bool want_rs2 = sel && strcmp(sel, "rs2") == 0 &&
                             n % 128 == 0;
             // rs8 (8-wave 128x64 tile) measures ~equal to rs (1.23 vs

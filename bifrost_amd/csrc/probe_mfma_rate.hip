@@ -10,6 +10,8 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <cstdlib>
+#include <ctime>
 
 typedef int v4i __attribute__((ext_vector_type(4)));
 
@@ -282,6 +284,8 @@ void mfma_tr8_kernel(const int* __restrict__ in, long* __restrict__ cyc,
     }
 }
 
+bool g_zero_data = false;
+
 template <typename K>
 void run(const char* name, K kern, int blocks, int iter, int nacc,
          int ldsbytes = 0) {
@@ -291,13 +295,27 @@ void run(const char* name, K kern, int blocks, int iter, int nacc,
     (void)hipMalloc(&in, 4096);
     (void)hipMalloc(&cyc, blocks * sizeof(long));
     (void)hipMalloc(&sink, blocks * sizeof(int));
-    (void)hipMemset(in, 1, 4096);
+    if (g_zero_data) {
+        (void)hipMemset(in, 1, 4096);
+    } else {
+        unsigned char rnd[4096];
+        unsigned s = 12345;
+        for (int i = 0; i < 4096; ++i) {
+            s = s * 1664525u + 1013904223u;
+            rnd[i] = (unsigned char)(s >> 24);
+        }
+        (void)hipMemcpy(in, rnd, 4096, hipMemcpyHostToDevice);
+    }
     hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), ldsbytes, 0, in, cyc,
                        sink, iter);
     (void)hipDeviceSynchronize();
+    timespec w0, w1;
+    clock_gettime(CLOCK_MONOTONIC, &w0);
     hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), ldsbytes, 0, in, cyc,
                        sink, iter);
     (void)hipDeviceSynchronize();
+    clock_gettime(CLOCK_MONOTONIC, &w1);
+    double wall = (w1.tv_sec - w0.tv_sec) + 1e-9 * (w1.tv_nsec - w0.tv_nsec);
     long* h = new long[blocks];
     (void)hipMemcpy(h, cyc, blocks * sizeof(long), hipMemcpyDeviceToHost);
     long mx = 0, mn = (long)1e18;
@@ -306,14 +324,25 @@ void run(const char* name, K kern, int blocks, int iter, int nacc,
         if (h[i] < mn) mn = h[i];
     }
     double per = (double)mx / ((double)iter * nacc);
-    printf("%-28s blocks=%4d nacc=%3d  cyc/MFMA max %.2f min %.2f\n", name,
-           blocks, nacc, per, (double)mn / ((double)iter * nacc));
+    // chip-wide sustained rate from wall (valid for the non-OUTER
+    // kernels where total equiv-MFMAs = blocks*4*iter*nacc; OUTER
+    // variants are 20x this — noted per row)
+    double tot = (double)blocks * 4.0 * iter * nacc;
+    double tops = tot * 32768.0 / wall / 1e12;
+    printf("%-28s blocks=%4d nacc=%3d  cyc/MFMA max %.2f min %.2f  "
+           "wall %.2f ms (%.0f equiv-TOPS)\n",
+           name, blocks, nacc, per, (double)mn / ((double)iter * nacc),
+           wall * 1e3, tops);
     delete[] h;
     (void)hipFree(in); (void)hipFree(cyc); (void)hipFree(sink);
 }
 
 int main() {
     int iter = 2000;
+    // Random operand data by default: low-toggle (memset-1) operands let
+    // the chip clock ~15% higher and overstate sustainable rates
+    // (PROBE_ZERO_DATA=1 restores the old behaviour for comparison).
+    g_zero_data = getenv("PROBE_ZERO_DATA") != nullptr;
     // pure MFMA chains
     run("occ1 nacc64 (rs2 shape)", mfma_rate_kernel<64, 1>, 256, iter, 64);
     run("occ1 nacc16", mfma_rate_kernel<16, 1>, 256, iter, 16);
@@ -323,6 +352,9 @@ int main() {
     run("occ4 nacc8", mfma_rate_kernel<8, 4>, 1024, iter, 8);
     // diverse 8x8 operand pairs (the rs2 burst register pattern)
     run("pairs occ1 (rs2 burst)", mfma_pairs_kernel<1>, 256, iter, 64);
+    // SUSTAINED power envelope: long full-chip burst on random data
+    run("SUSTAINED pairs 10x", mfma_pairs_kernel<1>, 256, 20000, 64);
+    run("SUSTAINED fixed 10x", mfma_rate_kernel<64, 1>, 256, 20000, 64);
     run("env base (pairs occ1)", mfma_pairs_env_kernel<0, 0, 1>, 256,
         iter, 64);
     run("env +lds69k", mfma_pairs_env_kernel<1, 0, 1>, 256, iter, 64,
