@@ -7,8 +7,12 @@
 // Space-aware: the buffer may live in HIP device memory; ghost fix-up
 // copies go through bfMemcpy on the calling thread's stream.
 //
-// Supported: nringlet == 1 (every block on the hot path uses flat frames;
-// multi-ringlet returns BF_STATUS_UNSUPPORTED_SHAPE).
+// Multi-ringlet (round 2): a ring with nringlet N holds N parallel
+// lanes; physical layout is N slabs of (capacity + ghost) bytes, so a
+// span's data pointer is ringlet 0's window and the ringlet stride is
+// capacity + ghost (what ring.py/ring2.py build their views from,
+// reference src/bifrost/ring.h:114 semantics).  Changing nringlet is
+// only allowed while the ring is empty (no data, no sequences).
 
 #include <bifrost/memory.h>
 #include <bifrost/ring.h>
@@ -56,6 +60,7 @@ struct BFring_impl {
     char* buf = nullptr;
     BFsize capacity = 0;   // bytes per ringlet (logical window)
     BFsize ghost = 0;      // contiguous-span bytes duplicated past the end
+    BFsize nringlet = 1;   // parallel lanes; slab stride = capacity + ghost
 
     bool writing_begun = false;
     bool writing_ended = false;
@@ -95,8 +100,13 @@ struct BFring_impl {
         BFsize pos = abs % capacity;
         if (pos + size <= capacity) return BF_STATUS_SUCCESS;
         BFsize overhang = pos + size - capacity;
-        BFstatus st = bfMemcpy(buf + capacity, space, buf, space, overhang);
-        if (st != BF_STATUS_SUCCESS) return st;
+        BFsize slab = capacity + ghost;
+        for (BFsize r = 0; r < nringlet; ++r) {
+            char* base = buf + r * slab;
+            BFstatus st = bfMemcpy(base + capacity, space, base, space,
+                                   overhang);
+            if (st != BF_STATUS_SUCCESS) return st;
+        }
         return ghost_fence();
     }
     // Propagate bytes written into the ghost area back to the buffer start
@@ -105,8 +115,13 @@ struct BFring_impl {
         BFsize pos = abs % capacity;
         if (pos + size <= capacity) return BF_STATUS_SUCCESS;
         BFsize overhang = pos + size - capacity;
-        BFstatus st = bfMemcpy(buf, space, buf + capacity, space, overhang);
-        if (st != BF_STATUS_SUCCESS) return st;
+        BFsize slab = capacity + ghost;
+        for (BFsize r = 0; r < nringlet; ++r) {
+            char* base = buf + r * slab;
+            BFstatus st = bfMemcpy(base, space, base + capacity, space,
+                                   overhang);
+            if (st != BF_STATUS_SUCCESS) return st;
+        }
         return ghost_fence();
     }
     // Ghost fix-up copies on device-space rings are enqueued async on the
@@ -179,13 +194,21 @@ BFstatus bfRingDestroy(BFring ring) {
 BFstatus bfRingResize(BFring ring, BFsize contiguous_bytes,
                       BFsize capacity_bytes, BFsize nringlet) {
     BF_ASSERT(ring, BF_STATUS_INVALID_HANDLE);
-    BF_ASSERT(nringlet == 1, BF_STATUS_UNSUPPORTED_SHAPE);
+    BF_ASSERT(nringlet >= 1, BF_STATUS_INVALID_ARGUMENT);
     BF_TRY_RETURN({
         std::unique_lock<std::mutex> lk(ring->mutex);
+        // Growing/shrinking the ringlet count is only defined while the
+        // ring holds no data (re-laning live bytes has no meaning).
+        BF_THROW_IF(nringlet != ring->nringlet &&
+                        (ring->head > 0 || !ring->sequences.empty()),
+                    BF_STATUS_UNSUPPORTED_SHAPE);
+        BFsize new_nringlet = nringlet;
         BFsize new_ghost = std::max(ring->ghost, contiguous_bytes);
         BFsize new_cap = std::max(ring->capacity,
                                   std::max(capacity_bytes, new_ghost));
-        if (new_ghost == ring->ghost && new_cap == ring->capacity) return BF_STATUS_SUCCESS;
+        if (new_ghost == ring->ghost && new_cap == ring->capacity &&
+            new_nringlet == ring->nringlet)
+            return BF_STATUS_SUCCESS;
         // Wait until no span is open: live spans hold raw pointers into the
         // old buffer, so reallocating under them is a use-after-free.  New
         // reserve/acquire calls block while nrealloc_pending > 0.
@@ -198,35 +221,44 @@ BFstatus bfRingResize(BFring ring, BFsize contiguous_bytes,
         new_ghost = std::max(ring->ghost, contiguous_bytes);
         new_cap = std::max(ring->capacity,
                            std::max(capacity_bytes, new_ghost));
-        if (new_ghost == ring->ghost && new_cap == ring->capacity) {
+        if (new_ghost == ring->ghost && new_cap == ring->capacity &&
+            new_nringlet == ring->nringlet) {
             ring->cv.notify_all();
             return BF_STATUS_SUCCESS;
         }
         char* new_buf = nullptr;
-        BF_THROW_IF(bfMalloc((void**)&new_buf, new_cap + new_ghost,
+        BF_THROW_IF(bfMalloc((void**)&new_buf,
+                             new_nringlet * (new_cap + new_ghost),
                              ring->space) != BF_STATUS_SUCCESS,
                     BF_STATUS_MEM_ALLOC_FAILED);
         if (ring->buf && ring->head > 0) {
-            // Re-place live bytes [tail, head) at their new positions.
-            BFoffset t = ring->tail();
-            BFoffset h = ring->head;
-            // Copy in chunks that are contiguous in BOTH old and new layout.
-            BFoffset o = t;
-            while (o < h) {
-                BFsize old_pos = o % ring->capacity;
-                BFsize new_pos = o % new_cap;
-                BFsize n = std::min((BFsize)(h - o),
-                                    std::min(ring->capacity - old_pos,
-                                             new_cap - new_pos));
-                bfMemcpy(new_buf + new_pos, ring->space,
-                         ring->buf + old_pos, ring->space, n);
-                o += n;
+            // Re-place live bytes [tail, head) at their new positions,
+            // per ringlet (nringlet unchanged here by the gate above).
+            BFsize old_slab = ring->capacity + ring->ghost;
+            BFsize new_slab = new_cap + new_ghost;
+            for (BFsize rl = 0; rl < ring->nringlet; ++rl) {
+                BFoffset t = ring->tail();
+                BFoffset h = ring->head;
+                // Chunks contiguous in BOTH old and new layout.
+                BFoffset o = t;
+                while (o < h) {
+                    BFsize old_pos = o % ring->capacity;
+                    BFsize new_pos = o % new_cap;
+                    BFsize n = std::min((BFsize)(h - o),
+                                        std::min(ring->capacity - old_pos,
+                                                 new_cap - new_pos));
+                    bfMemcpy(new_buf + rl * new_slab + new_pos, ring->space,
+                             ring->buf + rl * old_slab + old_pos,
+                             ring->space, n);
+                    o += n;
+                }
             }
         }
         if (ring->buf) bfFree(ring->buf, ring->space);
         ring->buf = new_buf;
         ring->capacity = new_cap;
         ring->ghost = new_ghost;
+        ring->nringlet = new_nringlet;
         ring->cv.notify_all();
     });
 }
@@ -321,7 +353,9 @@ BFstatus bfRingSequenceBegin(BFwsequence* sequence, BFring ring,
                              BFsize header_size, const void* header,
                              BFsize nringlet, BFoffset offset_from_head) {
     BF_ASSERT(sequence && ring, BF_STATUS_INVALID_POINTER);
-    BF_ASSERT(nringlet == 1, BF_STATUS_UNSUPPORTED_SHAPE);
+    BF_ASSERT(nringlet >= 1, BF_STATUS_INVALID_ARGUMENT);
+    // The ring's physical lanes must cover the sequence's (resize first).
+    BF_ASSERT(nringlet <= ring->nringlet, BF_STATUS_UNSUPPORTED_SHAPE);
     BF_TRY_RETURN({
         std::lock_guard<std::mutex> lk(ring->mutex);
         auto seq = std::make_shared<Sequence>();
@@ -677,8 +711,11 @@ BFstatus bfRingSpanGetInfo(BFspan span, BFspan_info* info) {
     info->ring = ring;
     info->data = ring->ptr_at(sp->begin);
     info->size = sp->size;
-    info->stride = ring->capacity;
-    info->nringlet = 1;
+    // ringlet stride = the per-lane slab (capacity + ghost); read spans
+    // report their sequence's lane count, write spans the ring's.
+    info->stride = ring->capacity + ring->ghost;
+    info->nringlet = sp->is_write ? ring->nringlet
+                                  : ((BFrspan_impl*)sp)->rseq->seq->nringlet;
     if (sp->is_write) {
         info->offset = (BFsize)sp->begin;
     } else {

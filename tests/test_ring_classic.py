@@ -160,3 +160,66 @@ def test_classic_gulp_larger_than_ring_grows_capacity():
         got = [s.data.tobytes() for s in iseq.read(1024)]
     t.join()
     assert got == [payload.tobytes()]
+
+
+def test_multi_ringlet_roundtrip():
+    """Round 2: nringlet > 1 — N parallel lanes sharing offsets; span
+    data is ringlet-strided (stride = capacity + ghost).  Writes fill
+    each lane with a distinct pattern; reads must see them laned, also
+    across the wrap (per-ringlet ghost fix-up)."""
+    import threading
+
+    import bifrost_amd.ring as bring
+
+    NRINGLET, GULP, NGULP = 4, 32, 12
+    ring = bring.Ring(name="mrl")
+    ring.resize(GULP, GULP * 4, nringlet=NRINGLET)
+
+    def writer():
+        with ring.begin_writing() as ow:
+            with ow.begin_sequence(name="mr", time_tag=1,
+                                   nringlet=NRINGLET) as oseq:
+                for g in range(NGULP):
+                    with oseq.reserve(GULP) as wspan:
+                        v = wspan.data_view(np.uint8)
+                        assert v.shape == (NRINGLET, GULP)
+                        for r in range(NRINGLET):
+                            v[r, :] = (np.arange(GULP) + 64 * r + g) % 256
+
+    t = threading.Thread(target=writer)
+    t.start()
+    seen = 0
+    with ring.open_earliest_sequence(guarantee=True) as iseq:
+        for g, span in enumerate(iseq.read(GULP)):
+            v = span.data_view(np.uint8)
+            assert v.shape == (NRINGLET, GULP)
+            for r in range(NRINGLET):
+                np.testing.assert_array_equal(
+                    v[r], (np.arange(GULP) + 64 * r + g) % 256)
+            seen += 1
+            if seen == NGULP:
+                break
+    t.join()
+    assert seen == NGULP
+
+
+def test_multi_ringlet_resize_rules():
+    import bifrost_amd.ring as bring
+    from bifrost_amd.libbifrost import _bf
+
+    ring = bring.Ring(name="mrl2")
+    ring.resize(16, 64, nringlet=3)
+    # changing lanes while empty is fine
+    ring.resize(16, 64, nringlet=2)
+    with ring.begin_writing() as ow:
+        with ow.begin_sequence(name="s", nringlet=2) as oseq:
+            with oseq.reserve(16) as wspan:
+                wspan.data_view(np.uint8)[...] = 7
+    # ring now holds data: changing nringlet must be refused
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        ring.resize(16, 64, nringlet=5)
+    # sequence lanes may not exceed the ring's
+    with _pytest.raises(RuntimeError):
+        with ring.begin_writing() as ow:
+            ow.begin_sequence(name="s2", nringlet=8)
