@@ -289,24 +289,57 @@ def run_c5(args):
         def on_data(self, ispan):
             self.nframe += ispan.nframe
 
+    stages = os.environ.get("BIFROST_C5_STAGES", "full")
+
     def run_pipeline(ngulp):
         drain = []
         with Pipeline() as pipe:
             src = DeviceVoltageSource(ngulp)
-            beam = BeamformBlock(src, w)
-            fine = views.split_axis(beam, 0, NFINE, label="fine_time")
-            spec = bf.blocks.fft(fine, axes="fine_time")
-            pwr = bf.blocks.detect(spec, mode="scalar")
-            acc = bf.blocks.accumulate(pwr, TGULP // NFINE)
-            drain.append(DrainBlock(acc))
+            if stages == "src":
+                tail = src
+            elif stages == "beam":
+                tail = BeamformBlock(src, w)
+            elif stages == "fft":
+                beam = BeamformBlock(src, w)
+                fine = views.split_axis(beam, 0, NFINE, label="fine_time")
+                tail = bf.blocks.fft(fine, axes="fine_time")
+            elif stages == "detect":
+                beam = BeamformBlock(src, w)
+                fine = views.split_axis(beam, 0, NFINE, label="fine_time")
+                spec = bf.blocks.fft(fine, axes="fine_time")
+                tail = bf.blocks.detect(spec, mode="scalar")
+            else:
+                beam = BeamformBlock(src, w)
+                fine = views.split_axis(beam, 0, NFINE, label="fine_time")
+                spec = bf.blocks.fft(fine, axes="fine_time")
+                pwr = bf.blocks.detect(spec, mode="scalar")
+                tail = bf.blocks.accumulate(pwr, TGULP // NFINE)
+            drain.append(DrainBlock(tail))
             t0 = time.perf_counter()
             pipe.run()
             torch.cuda.synchronize()
             dt = time.perf_counter() - t0
         return dt, drain[0].nframe
 
+    def dump_block_perf(tag):
+        # per-block last-span process_time from the proclog tree
+        import glob as _glob
+        base = os.path.join(
+            os.environ.get("BIFROST_PROCLOG_DIR", "/dev/shm/bifrost_amd"),
+            str(os.getpid()))
+        for path in sorted(_glob.glob(os.path.join(base, "*", "perf"))):
+            try:
+                with open(path) as f:
+                    print("PERF[%s] %s: %s"
+                          % (tag, os.path.basename(os.path.dirname(path)),
+                             f.read().strip()), file=sys.stderr)
+            except OSError:
+                pass
+
     run_pipeline(max(2, args.warmup))           # spin-up: JIT, FFT plans
     dt, nspec = run_pipeline(args.steps)
+    if os.environ.get("BIFROST_C5_PERF"):
+        dump_block_perf(stages)
     samples = args.steps * TGULP * NCHAN_C5      # (time, chan) samples
     value = samples / dt / 1e9
     # The dominant kernel is the bf16-split MFMA beamformer: flops =
