@@ -58,10 +58,21 @@ def raw_memset(ptr, space, value, count):
 
 def _get_space(arr):
     """Space of a numpy/bifrost array ('system' for plain numpy —
-    reference memory.py:_get_space)."""
+    reference memory.py:_get_space).
+
+    Round-2 fix: bifrost_amd ndarrays carry their space in `.bf.space`
+    (numpy's flags object has no custom keys, so the flags lookup always
+    fell through to 'system' — device-device copies then took the HOST
+    std::memcpy path through BAR-mapped HBM at ~50 MB/s: correct bytes,
+    ~30000x slow; gpu_c5_probe.py measurements)."""
+    bfmeta = getattr(arr, "bf", None)
+    if bfmeta is not None:
+        space = getattr(bfmeta, "space", None)
+        if space:
+            return str(space)
     try:
         return arr.flags["SPACE"]
-    except (AttributeError, KeyError):
+    except (AttributeError, KeyError, TypeError):
         return "system"
 
 
