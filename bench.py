@@ -385,7 +385,41 @@ def run_c5(args):
         },
         "drained_frames": nspec,
     }
+    if not getattr(args, "skip_cpu_baseline", False):
+        result["cpu_baseline"] = c5_cpu_baseline(w, max_seconds=10.0)
     print(json.dumps(result))
+
+
+def c5_cpu_baseline(w, max_seconds=10.0):
+    """The C5 beamform stage's numpy/BLAS restatement timed on host
+    cores: per channel, W[nbeam, n] @ X^H[n, T] cgemm at the full C5
+    per-channel shape (the FFT/detect stages are negligible beside it on
+    CPU).  Bounded sample, scaled to (t,chan)-samples/sec."""
+    nbeam, nchan, n = w.shape
+    T = 256
+    rng = np.random.RandomState(5)
+    xc = np.ascontiguousarray(
+        (rng.randint(-7, 8, size=(T, n)) +
+         1j * rng.randint(-7, 8, size=(T, n))).astype(np.complex64).T)
+    _ = w[:, 0] @ xc[:, :32]
+    t0 = time.perf_counter()
+    done = 0
+    while done < nchan:
+        y = w[:, done % nchan] @ xc
+        done += 1
+        if time.perf_counter() - t0 > max_seconds:
+            break
+    dt = time.perf_counter() - t0
+    assert y.shape == (nbeam, T)
+    return {
+        "value": done * T / dt / 1e9,
+        "unit": "Gsamp/s",
+        "cores": physical_cores(),
+        "kind": "port",
+        "sample": "numpy/BLAS beamform, %d x full C5 per-channel cgemm "
+                  "(nbeam=%d, n=%d, T=%d) in %.1fs"
+                  % (done, nbeam, n, T, dt),
+    }
 
 
 def main():
