@@ -1619,7 +1619,13 @@ void cherk_ci8_mfma32_rs4_kernel(long n, long k, long nbatch, float alpha,
 // the round-2 zero-VALU addressing: 8 instructions per wave-slab
 // (vs rs's 16), 64 accumulator regs, ~150-reg footprint -> 3 waves/SIMD
 // with 4-WG LDS headroom (both strips [64][144]).
-#define RS5_ROW 144
+// 160-B rows (round-2 fix): the first rs5 used 144-B rows, whose b128
+// staging writes collide 2-way within each write group (measured 48% of
+// LDS cycles as SQ_LDS_BANK_CONFLICT, profiles/round2_rs5_pmc.md).  At
+// 160 B both access patterns are conflict-free: tr8 gather banks
+// (40r + {0,2,4,6}) mod 64 all distinct per group; b128 write banks
+// (8r + 8e + 4h) mod 32 all distinct.  40 KB total keeps 4 WGs/CU.
+#define RS5_ROW 160
 #define RS5_STRIP (64 * RS5_ROW)
 #define RS5_BUF (2 * RS5_STRIP)
 template <int SCHED>
