@@ -2562,7 +2562,9 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
             bool want_rs8 = sel8 && strcmp(sel8, "rs8") == 0;
             if (al16 && want_rs5) {
                 const char* schenv5 = getenv("BIFROST_CHERK_SCHED");
-                int sched5 = schenv5 ? atoi(schenv5) : 0;
+                // sched 1 (burst first, then writes+loads) wins the
+                // same-box ABAB consistently (r2_ab9/r2_ab11)
+                int sched5 = schenv5 ? atoi(schenv5) : 1;
                 if (sched5 == 1)
                     hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<1>, grid,
                                        dim3(256), 0, stream, n, k, nbatch,
@@ -2831,9 +2833,13 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
 // regardless of registers) and PRE=0 (no cross-slab register prefetch;
 // the third wave covers staging latency instead).  Default stays the
 // round-1 prefetch form until the same-box A/B decides.
+// Round-2 default: the 3-wave variant (JT=1, PRE=0) — its worst-case
+// rate beats the 2-wave prefetch form on every box measured
+// (r2_beam1/r2_beam2: ci8 363-393 vs 309-375 TF); BIFROST_BEAM=mfma2
+// restores the round-1 kernel.
 #define BEAM_MFMA_ONE(NBT, XTV, WTV)                                          \
     do {                                                                      \
-        if (beam_sel && strcmp(beam_sel, "mfma3") == 0) {                     \
+        if (!(beam_sel && strcmp(beam_sel, "mfma2") == 0)) {                     \
             dim3 mgrid3(cap_grid(nn / 64, 4096), cap_grid(nbatch, 65535));    \
             hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV, 1, 0>),   \
                                mgrid3, dim3(256), 0, stream, nn, k, nbatch,   \
