@@ -29,7 +29,8 @@ def main():
     bf_device.set_stream(torch.cuda.current_stream().cuda_stream)
     la = LinAlg()
 
-    variants = [("mfma2", None), ("mfma3", "mfma3")]
+    # round-2: the 3-wave kernel is the default; "mfma2" forces round-1
+    variants = [("mfma2", "mfma2"), ("mfma3", None)]
 
     def set_env(v):
         os.environ.pop("BIFROST_BEAM", None)
