@@ -267,8 +267,8 @@ class TestMapCI4:
     def test_map_write_ci4(self):
         # quantize through map: b(ci4) = a(ci8)
         vals = np.zeros((3,), dtype=bf.DataType.ci8)
-        vals["re_im"]["re"] = [1, -3, 7]
-        vals["re_im"]["im"] = [2, -4, -1]
+        vals["re"] = [1, -3, 7]
+        vals["im"] = [2, -4, -1]
         a = bf.asarray(bf.ndarray(vals), space="cuda")
         b = bf.ndarray(shape=(3,), dtype="ci4", space="cuda")
         bf.map("b = a", {"b": b, "a": a})
