@@ -313,9 +313,17 @@ def run_c5(args):
             else:
                 beam = BeamformBlock(src, w)
                 fine = views.split_axis(beam, 0, NFINE, label="fine_time")
-                spec = bf.blocks.fft(fine, axes="fine_time")
-                pwr = bf.blocks.detect(spec, mode="scalar")
-                tail = bf.blocks.accumulate(pwr, TGULP // NFINE)
+                nspec = TGULP // NFINE
+                # one span per source gulp through the tail blocks: each
+                # span carries ~1 ms of pipeline overhead (round-2 C5
+                # bisect), so fft/detect gulp the full spectra batch and
+                # accumulate takes the whole window as one bfReduce
+                spec = bf.blocks.fft(fine, axes="fine_time",
+                                     gulp_nframe=nspec)
+                pwr = bf.blocks.detect(spec, mode="scalar",
+                                       gulp_nframe=nspec)
+                tail = bf.blocks.accumulate(pwr, nspec,
+                                            gulp_nframe=nspec)
             drain.append(DrainBlock(tail))
             t0 = time.perf_counter()
             pipe.run()
