@@ -174,11 +174,16 @@ def test_multi_ringlet_roundtrip():
     NRINGLET, GULP, NGULP = 4, 32, 12
     ring = bring.Ring(name="mrl")
     ring.resize(GULP, GULP * 4, nringlet=NRINGLET)
+    # a reader's guarantee protects data only from its OPEN onwards
+    # (reference semantics): gate the writer on the reader being open so
+    # the 4-gulp window can't lap it under suite load
+    reader_open = threading.Event()
 
     def writer():
         with ring.begin_writing() as ow:
             with ow.begin_sequence(name="mr", time_tag=1,
                                    nringlet=NRINGLET) as oseq:
+                assert reader_open.wait(timeout=30)
                 for g in range(NGULP):
                     with oseq.reserve(GULP) as wspan:
                         v = wspan.data_view(np.uint8)
@@ -190,6 +195,7 @@ def test_multi_ringlet_roundtrip():
     t.start()
     seen = 0
     with ring.open_earliest_sequence(guarantee=True) as iseq:
+        reader_open.set()
         for g, span in enumerate(iseq.read(GULP)):
             v = span.data_view(np.uint8)
             assert v.shape == (NRINGLET, GULP)
