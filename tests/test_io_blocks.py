@@ -333,3 +333,26 @@ class TestDadaFile:
             pipe.run()
         got = np.concatenate(out, axis=0)
         np.testing.assert_array_equal(got, data)  # spans both files
+
+
+@pytest.mark.gpu
+def test_accumulate_device_whole_window_reduce():
+    """Round 2: device-space accumulate with gulp_nframe == window goes
+    through ONE bfReduce per output instead of per-frame map launches
+    (the C5 pipeline's configuration)."""
+    import bifrost_amd as bf
+    from tests.test_pipeline_cpu import NumpySourceBlock, CollectBlock
+
+    rng = np.random.RandomState(3)
+    data = rng.standard_normal((24, 5, 7)).astype(np.float32)
+    out = []
+    with bf.Pipeline() as pipe:
+        src = NumpySourceBlock([data], gulp_nframe=8)
+        dev = bf.blocks.copy(src, space="cuda")
+        acc = bf.blocks.accumulate(dev, 8, gulp_nframe=8)
+        host = bf.blocks.copy(acc, space="system")
+        CollectBlock(host, out)
+        pipe.run()
+    got = np.concatenate(out, axis=0)
+    gold = data.reshape(3, 8, 5, 7).sum(axis=1)
+    np.testing.assert_allclose(got, gold, rtol=1e-5, atol=1e-5)

@@ -3,6 +3,7 @@
 
 import threading
 
+import pytest
 import numpy as np
 
 from bifrost_amd.ring import Ring
@@ -229,3 +230,36 @@ def test_multi_ringlet_resize_rules():
     with _pytest.raises(RuntimeError):
         with ring.begin_writing() as ow:
             ow.begin_sequence(name="s2", nringlet=8)
+
+
+@pytest.mark.gpu
+def test_multi_ringlet_device_ring():
+    """Multi-ringlet lanes on a cuda-space ring: per-ringlet ghost
+    fix-up runs device-side copies (with the round-2 stream fence)."""
+    import bifrost_amd.ring as bring
+    from bifrost_amd import memory as bf_memory
+    import bifrost_amd as bf
+
+    NR, GULP = 3, 64
+    ring = bring.Ring(name="mrl_dev", space="cuda")
+    ring.resize(GULP, GULP * 3, nringlet=NR)
+    pats = [(np.arange(GULP, dtype=np.uint8) + 17 * r) % 251
+            for r in range(NR)]
+    with ring.begin_writing() as ow:
+        with ow.begin_sequence(name="d", nringlet=NR) as oseq:
+            # 5 gulps on a 3-gulp window forces wrap + ghost flush
+            for g in range(5):
+                with oseq.reserve(GULP) as wspan:
+                    v = wspan.data_view(np.uint8)
+                    assert v.shape == (NR, GULP)
+                    for r in range(NR):
+                        host = bf.ndarray((pats[r] + g) % 251)
+                        bf_memory.memcpy(v[r], host)
+    # v[r] of the LAST gulp should still hold its pattern
+    with ring.open_latest_sequence(guarantee=True) as iseq:
+        spans = iseq.read(GULP, begin=4 * GULP)
+        span = next(iter(spans))
+        v = span.data_view(np.uint8)
+        for r in range(NR):
+            got = np.asarray(bf.ndarray(v[r]).copy("system"))
+            np.testing.assert_array_equal(got, (pats[r] + 4) % 251)
