@@ -1408,8 +1408,12 @@ void cherk_ci8_mfma32_rs3_kernel(long n, long k, long nbatch, float alpha,
 // complex over a rectangular-triangular tile map (row blocks of 128,
 // col blocks of 64, J <= 2I+1 — the rs8 map).  Fragment gather and
 // epilogue layout identical to rs3 (probe-verified).
-#define RS4_JROW 144
-#define RS4_ISTRIP (64 * RS2_ROW)
+// Conflict-free strides (same bank argument as rs5's 160-B fix):
+// I rows 288 B (banks 8r+{0,2,4,6} mod 64 distinct; writes 8r+8e+4h
+// mod 32 distinct), J rows 160 B.
+#define RS4_IROW 288
+#define RS4_JROW 160
+#define RS4_ISTRIP (64 * RS4_IROW)
 #define RS4_JSTRIP (64 * RS4_JROW)
 template <int SCHED>
 __global__ __launch_bounds__(256)
@@ -1419,7 +1423,7 @@ void cherk_ci8_mfma32_rs4_kernel(long n, long k, long nbatch, float alpha,
                                  long a_b, float beta, f2* __restrict__ c,
                                  long c_row, long c_b, long nti,
                                  long ntiles) {
-    __shared__ signed char ldsI[2][64][RS2_ROW];
+    __shared__ signed char ldsI[2][64][RS4_IROW];
     __shared__ signed char ldsJ[2][64][RS4_JROW];
     int tid = threadIdx.x;
     int lane = tid & 63;
@@ -1436,11 +1440,11 @@ void cherk_ci8_mfma32_rs4_kernel(long n, long k, long nbatch, float alpha,
     int st_row = tt >> 1;
     int st_h = tt & 1;
 
-    const signed char* rdI0 = &ldsI[0][0][0] + row32 * RS2_ROW + colb32
+    const signed char* rdI0 = &ldsI[0][0][0] + row32 * RS4_IROW + colb32
                             + 128 * wr;
     const signed char* rdJ0 = &ldsJ[0][0][0] + row32 * RS4_JROW + colb32
                             + 64 * wc;
-    signed char* wrI0 = &ldsI[0][0][0] + st_row * RS2_ROW + 16 * st_h;
+    signed char* wrI0 = &ldsI[0][0][0] + st_row * RS4_IROW + 16 * st_h;
     signed char* wrJ0 = &ldsJ[0][0][0] + st_row * RS4_JROW + 16 * st_h;
 
     long total = 8 * ntiles * ((nbatch + 7) / 8);
@@ -1506,10 +1510,10 @@ void cherk_ci8_mfma32_rs4_kernel(long n, long k, long nbatch, float alpha,
             }
         };
         auto fragI = [&](const signed char* base, int h, int cc) {
-            const signed char* p = base + (32 * h) * RS2_ROW + 32 * cc;
+            const signed char* p = base + (32 * h) * RS4_IROW + 32 * cc;
             v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
             v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
-                (lds_v2i)(p + 8 * RS2_ROW));
+                (lds_v2i)(p + 8 * RS4_IROW));
             return v4i{lo[0], lo[1], hi[0], hi[1]};
         };
         auto fragJ = [&](const signed char* base, int h, int cc) {
@@ -1721,9 +1725,35 @@ void cherk_ci8_mfma32_rs5_kernel(long n, long k, long nbatch, float alpha,
                     }
             }
         };
+        // SCHED 13 diagnostic: fixed fragments read once per tile
+        v4i ffa[4], ffb[4];
+        if (SCHED == 13) {
+#pragma unroll
+            for (int h = 0; h < 2; ++h) {
+                ffb[2 * h] = frag(rdJ0, h, 0);
+                ffb[2 * h + 1] = frag(rdJ0, h, 1);
+                ffa[2 * h] = frag(rdI0, h, 0);
+                ffa[2 * h + 1] = frag(rdI0, h, 1);
+            }
+        }
+        auto burst_fixed = [&]() {
+#pragma unroll
+            for (int h = 0; h < 2; ++h)
+#pragma unroll
+                for (int ta = 0; ta < 2; ++ta)
+#pragma unroll
+                    for (int tb = 0; tb < 2; ++tb)
+                        acc[ta][tb] = __builtin_amdgcn_mfma_i32_32x32x32_i8(
+                            ffa[2 * h + ta], ffb[2 * h + tb], acc[ta][tb],
+                            0, 0, 0);
+        };
         auto step = [&](int buf, int s, int nslab) {
             const signed char* bI = rdI0 + buf * RS5_BUF;
             const signed char* bJ = rdJ0 + buf * RS5_BUF;
+            if (SCHED == 13) {
+                if (!skip_all) burst_fixed();
+                return;
+            }
             if (SCHED == 0) {
                 if (s + 1 < nslab) {
                     write_slab(buf ^ 1);
@@ -1745,6 +1775,8 @@ void cherk_ci8_mfma32_rs5_kernel(long n, long k, long nbatch, float alpha,
         write_slab(0);
         if (nslab > 1) load_slab();
         __syncthreads();
+        long tt0 = 0;
+        if (SCHED >= 12) tt0 = __builtin_amdgcn_s_memtime();
         int s = 0;
         while (s < nslab) {
             step(0, s, nslab);
@@ -1752,6 +1784,11 @@ void cherk_ci8_mfma32_rs5_kernel(long n, long k, long nbatch, float alpha,
             if (s >= nslab) break;
             step(1, s, nslab);
             ++s;
+        }
+        if (SCHED >= 12) {
+            long tt1 = __builtin_amdgcn_s_memtime();
+            if (lane == 0)
+                ((long long*)c)[(blockIdx.x & 16383) * 4 + wave] = tt1 - tt0;
         }
 #pragma unroll
         for (int ta = 0; ta < 2; ++ta) {
@@ -2565,7 +2602,19 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 // sched 1 (burst first, then writes+loads) wins the
                 // same-box ABAB consistently (r2_ab9/r2_ab11)
                 int sched5 = schenv5 ? atoi(schenv5) : 1;
-                if (sched5 == 1)
+                if (sched5 == 12)  // diagnostic: k-loop cycle dump
+                    hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<12>, grid,
+                                       dim3(256), 0, stream, n, k, nbatch,
+                                       (float)alpha, (const signed char*)a,
+                                       a_k, a_b, (float)beta, (f2*)c, c_row,
+                                       c_b, ntiles);
+                else if (sched5 == 13)  // diagnostic: pure + cycle dump
+                    hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<13>, grid,
+                                       dim3(256), 0, stream, n, k, nbatch,
+                                       (float)alpha, (const signed char*)a,
+                                       a_k, a_b, (float)beta, (f2*)c, c_row,
+                                       c_b, ntiles);
+                else if (sched5 == 1)
                     hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<1>, grid,
                                        dim3(256), 0, stream, n, k, nbatch,
                                        (float)alpha, (const signed char*)a,
