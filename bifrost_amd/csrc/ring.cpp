@@ -148,6 +148,11 @@ struct BFrsequence_impl {
     std::shared_ptr<Sequence> seq;
     bool guarantee = false;
     BFoffset guard = 0;  // absolute offset this reader still needs
+    // Spans keep their reader alive across an early Close (round-2 fix:
+    // an abandoned read generator finalizing after its sequence closed
+    // released through a freed rseq — use-after-free).
+    int nspan_open = 0;
+    bool closed = false;
 };
 
 // Common span header: BFspan is either a write or a read span; the shared
@@ -473,12 +478,15 @@ BFstatus bfRingSequenceNext(BFrsequence sequence) {
 BFstatus bfRingSequenceClose(BFrsequence sequence) {
     BF_ASSERT(sequence, BF_STATUS_INVALID_HANDLE);
     BFring_impl* ring = sequence->ring;
+    bool defer;
     {
         std::lock_guard<std::mutex> lk(ring->mutex);
         ring->guaranteed_readers.erase(sequence);
         ring->cv.notify_all();
+        defer = sequence->nspan_open > 0;
+        if (defer) sequence->closed = true;  // last span release frees
     }
-    delete sequence;
+    if (!defer) delete sequence;
     return BF_STATUS_SUCCESS;
 }
 
@@ -657,6 +665,7 @@ BFstatus bfRingSpanAcquire(BFrspan* span, BFrsequence sequence,
             }
             sequence->guard = begin;
             ++ring->nread_open;
+            ++sequence->nspan_open;
             auto* rs = new BFrspan_impl();
             rs->is_write = false;
             rs->ring = ring;
@@ -675,13 +684,16 @@ BFstatus bfRingSpanRelease(BFrspan span) {
     BF_ASSERT(span, BF_STATUS_INVALID_HANDLE);
     BFrsequence_impl* rseq = span->rseq;
     BFring_impl* ring = rseq->ring;
+    bool free_rseq;
     {
         std::lock_guard<std::mutex> lk(ring->mutex);
         // Advance the reader's guard past this span (sequential-gulp model).
         rseq->guard = std::max(rseq->guard, span->begin + span->size);
         --ring->nread_open;
+        free_rseq = (--rseq->nspan_open == 0) && rseq->closed;
         ring->cv.notify_all();
     }
+    if (free_rseq) delete rseq;
     delete span;
     return BF_STATUS_SUCCESS;
 }

@@ -255,7 +255,10 @@ def test_multi_ringlet_device_ring():
                     for r in range(NR):
                         host = bf.ndarray((pats[r] + g) % 251)
                         bf_memory.memcpy(v[r], host)
-    # v[r] of the LAST gulp should still hold its pattern
+    # v[r] of the LAST gulp should still hold its pattern.  (An
+    # abandoned read generator is ALSO safe — bfRingSequenceClose defers
+    # the reader free to the last span release, tested by leaving
+    # `spans` un-closed here.)
     with ring.open_latest_sequence(guarantee=True) as iseq:
         spans = iseq.read(GULP, begin=4 * GULP)
         span = next(iter(spans))
