@@ -245,6 +245,11 @@ def test_multi_ringlet_device_ring():
     ring.resize(GULP, GULP * 3, nringlet=NR)
     pats = [(np.arange(GULP, dtype=np.uint8) + 17 * r) % 251
             for r in range(NR)]
+    # persistent host buffers: bfMemcpy H2D is async on the thread
+    # stream, so the source must outlive the call until the sync below
+    hosts = [[bf.ndarray((pats[r] + g) % 251) for r in range(NR)]
+             for g in range(5)]
+    from bifrost_amd import device as bf_device
     with ring.begin_writing() as ow:
         with ow.begin_sequence(name="d", nringlet=NR) as oseq:
             # 5 gulps on a 3-gulp window forces wrap + ghost flush
@@ -253,8 +258,8 @@ def test_multi_ringlet_device_ring():
                     v = wspan.data_view(np.uint8)
                     assert v.shape == (NR, GULP)
                     for r in range(NR):
-                        host = bf.ndarray((pats[r] + g) % 251)
-                        bf_memory.memcpy(v[r], host)
+                        bf_memory.memcpy(v[r], hosts[g][r])
+                    bf_device.stream_synchronize()
     # v[r] of the LAST gulp should still hold its pattern.  (An
     # abandoned read generator is ALSO safe — bfRingSequenceClose defers
     # the reader free to the last span release, tested by leaving
