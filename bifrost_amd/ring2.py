@@ -277,8 +277,11 @@ class ReadSequence(SequenceBase):
                 buffer_factor = 3
             buf_nframe = int(np.ceil(gulp_nframe * buffer_factor))
         tensor = self.tensor
+        # keep the ring's ringlet count (a reader growing the window must
+        # not try to re-lane a live multi-ringlet ring)
         return self._ring.resize(gulp_nframe * tensor["frame_nbyte"],
-                                 buf_nframe * tensor["frame_nbyte"])
+                                 buf_nframe * tensor["frame_nbyte"],
+                                 nringlet=tensor["nringlet"])
 
     @property
     def header(self):

@@ -204,3 +204,57 @@ def test_pipeline_accumulate_block_dtype_upconvert():
             + 1j * raw["im"].astype(np.float32))
     np.testing.assert_allclose(got[0], want[0:4].sum(axis=0))
     np.testing.assert_allclose(got[1], want[4:8].sum(axis=0))
+
+
+def test_pipeline_ringlet_axis():
+    """Round 2: a sequence with a ringlet axis (shape [2, -1, 4]) flows
+    through the ring2 pipeline — the ring allocates 2 lanes and span
+    views expose [ringlet, frame, ...] with the lane stride."""
+    data = np.arange(2 * 24 * 4, dtype=np.float32).reshape(2, 24, 4)
+
+    class RingletSource(SourceBlock):
+        def __init__(self, **kw):
+            super(RingletSource, self).__init__(["r"], 8, **kw)
+
+        def create_reader(self, name):
+            class _R(object):
+                pos = 0
+
+                def __enter__(self):
+                    return self
+
+                def __exit__(self, *a):
+                    return False
+            return _R()
+
+        def on_sequence(self, reader, name):
+            return [{
+                "name": "ringlet-seq",
+                "time_tag": 0,
+                "_tensor": {
+                    "dtype": "f32",
+                    "shape": [2, -1, 4],
+                    "labels": ["pol", "time", "d"],
+                    "scales": [None, [0, 1], None],
+                    "units": [None, None, None],
+                },
+                "gulp_nframe": 8,
+            }]
+
+        def on_data(self, reader, ospans):
+            ospan = ospans[0]
+            n = min(ospan.nframe, 24 - reader.pos)
+            if n <= 0:
+                return [0]
+            np.asarray(ospan.data)[:, :n] = \
+                data[:, reader.pos:reader.pos + n]
+            reader.pos += n
+            return [n]
+
+    out = []
+    with bf.Pipeline() as pipe:
+        src = RingletSource()
+        CollectBlock(src, out)
+        pipe.run()
+    got = np.concatenate(out, axis=1)
+    np.testing.assert_array_equal(got, data)
