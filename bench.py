@@ -178,9 +178,9 @@ def run_c5(args):
         TransformBlock
 
     NBEAM, NCHAN_C5, NFINE = 64, 512, 64
-    # 1024-frame gulps amortize the per-span pipeline overhead (~1 ms per
-    # block-stage handoff measured at TGULP=256, gpurun_out/c5_*.log)
-    TGULP = 1024
+    # 4096-frame gulps amortize the per-span pipeline overhead (~0.8 ms
+    # fixed per gulp measured at TGULP=1024; see profiles/round2_c5.md)
+    TGULP = 4096
     n = N
 
     bf_device.set_device(0)
