@@ -29,3 +29,17 @@ def test_transform_units():
     assert transform_units("s", 1) == "s"
     assert transform_units(None, 2) is None
     assert transform_units("s", 2) == "s^2"
+
+
+def test_traffic_manifest_fresh():
+    """Round 2: the committed roofline.traffic constant is
+    keyed by the kernel-source hash; this test goes red the moment
+    linalg.hip changes without a TCC re-profile + manifest update
+    (profiles/traffic_manifest.json), which is the intended forcing
+    function — re-profile, update the entry, and this passes again."""
+    import bench
+    t = bench.traffic_bytes_per_launch("xcorr_n512_c512_t4096")
+    assert t is not None and t > 1e9, (
+        "traffic manifest is stale for the current bifrost_amd/csrc/"
+        "linalg.hip — re-run the TCC pass and update "
+        "profiles/traffic_manifest.json")
