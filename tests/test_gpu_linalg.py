@@ -95,6 +95,49 @@ class TestCorrelatorKernel:
         # and on the odd-shape (non-MFMA) correlator path
         run_corr(linalg, 65, 17, 2, beta=0.5)
 
+    def test_alpha_scaling(self, linalg):
+        # alpha != 1 on the MFMA correlator epilogue (rs5 path via
+        # nstand 64 and the rs path via nstand 32), with and without
+        # beta — the whole sweep elsewhere runs alpha=1 only
+        np.random.seed(11)
+        for nstand in (32, 64):
+            x_shape = (128, 2, nstand * 2)
+            x8 = ((np.random.random(size=x_shape + (2,)) * 2 - 1) * 127) \
+                .astype(np.int8)
+            x = x8.astype(np.float32).view(np.complex64).reshape(x_shape)
+            xv = x.transpose(1, 0, 2)
+            gold = np.matmul(H(xv), xv)
+            tri = np.triu_indices(xv.shape[-1], 1)
+            gold[..., tri[0], tri[1]] = 0
+            xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                       .reshape(x_shape)), space="cuda")
+            b = bf.zeros_like(gold, space="cuda")
+            linalg.matmul(2.5, None, xb.transpose(1, 0, 2), 0, b)
+            np.testing.assert_allclose(np.asarray(b.copy("system")),
+                                       2.5 * gold, RTOL * 10, ATOL)
+            linalg.matmul(-0.5, None, xb.transpose(1, 0, 2), 1.0, b)
+            np.testing.assert_allclose(np.asarray(b.copy("system")),
+                                       2.0 * gold, RTOL * 10, ATOL * 10)
+
+    def test_alpha_beamform(self, linalg):
+        np.random.seed(12)
+        t, bm, s, c = 64, 16, 64, 2
+        ks = s * 2
+        x8 = ((np.random.random(size=(t, c, ks, 2)) * 2 - 1) * 127) \
+            .astype(np.int8)
+        x = x8.astype(np.float32).view(np.complex64).reshape(t, c, ks)
+        w = ((np.random.random((bm, c, ks, 2)) * 2 - 1)).astype(np.float32) \
+            .view(np.complex64).reshape(bm, c, ks)
+        gold = np.matmul(w.transpose(1, 0, 2), x.transpose(1, 2, 0))
+        xb = bf.asarray(bf.ndarray(x8.view(bf.DataType.ci8)
+                                   .reshape(t, c, ks)), space="cuda")
+        wb = bf.asarray(w, space="cuda")
+        out = bf.zeros_like(gold, space="cuda")
+        linalg.matmul(3.0, wb.transpose(1, 0, 2), xb.transpose(1, 2, 0),
+                      0, out)
+        np.testing.assert_allclose(np.asarray(out.copy("system")),
+                                   3.0 * gold, RTOL * 10, ATOL * 10)
+
     def test_ci8_minus_128(self, linalg):
         # The reference's own test excludes -128 (test_linalg.py:55); our
         # kernels accumulate exactly in i32, so -128 is handled with no
