@@ -599,7 +599,8 @@ def main():
             },
             "roofline": roofline,
         }
-        if not args.skip_cpu_baseline:
+        if not args.skip_cpu_baseline and world_size == 1:
+            # contract: the CPU-baseline leg runs on rank 0 at N=1 only
             result["cpu_baseline"] = cpu_baseline()
         print(json.dumps(result))
 
