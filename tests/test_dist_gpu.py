@@ -38,7 +38,8 @@ sys.path.insert(0, os.environ["BIFROST_REPO"])
 
 rank = int(os.environ["RANK"])
 world = int(os.environ["WORLD_SIZE"])
-dist.init_process_group("nccl", rank=rank, world_size=world)
+backend = os.environ.get("BIFROST_TEST_BACKEND", "nccl")
+dist.init_process_group(backend, rank=rank, world_size=world)
 torch.cuda.set_device(0)
 
 import bifrost_amd as bf
@@ -82,7 +83,7 @@ print("RANK%d_OK" % rank)
 """
 
 
-def _launch(world, port):
+def _launch(world, port, backend="nccl"):
     procs = []
     for rank in range(world):
         env = dict(os.environ)
@@ -92,6 +93,7 @@ def _launch(world, port):
             "RANK": str(rank),
             "WORLD_SIZE": str(world),
             "BIFROST_REPO": _REPO,
+            "BIFROST_TEST_BACKEND": backend,
             # dmabuf IPC (see environment contract); required for RCCL
             "HSA_ENABLE_IPC_MODE_LEGACY": "0",
         })
@@ -132,3 +134,17 @@ def test_rccl_two_ranks_one_gpu_probe():
                     "(Duplicate GPU detected) — N>1 needs a multi-GPU "
                     "node; see docstring")
     pytest.fail("unexpected failure:\n%s" % joined)
+
+
+def test_two_ranks_one_gpu_gloo_combine():
+    """The FULL 2-rank time-split path ON HARDWARE: both ranks run the
+    HIP cherk kernel on the one GPU and combine visibilities with a
+    world-2 all_reduce.  RCCL forbids co-resident ranks (see the probe
+    above), so the collective transport here is gloo — which accepts
+    CUDA tensors — leaving only the wire protocol different from the
+    N>1 production path.  Results must match the full-integration
+    oracle."""
+    procs, outs = _launch(2, 29575, backend="gloo")
+    for rank, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, "rank %d failed:\n%s" % (rank, out)
+        assert "RANK%d_OK" % rank in out, out
