@@ -14,8 +14,13 @@ run).  What CAN run here and does:
   * the 2-rank attempt, kept as a probe: it must either pass (future
     RCCL permitting co-residence) or fail with the known Duplicate-GPU
     signature, which the test records via skip.
-Combine-semantics at world>1 are covered by the gloo world-2 CPU tests
-(tests/test_dist_cpu.py), which share this code path."""
+  * the FULL 2-rank combine ON the GPU (test_two_ranks_one_gpu_gloo_
+    combine): both ranks run the HIP cherk kernel co-resident on the
+    one device and all_reduce the visibilities at world 2 — gloo
+    transport, since RCCL rejects co-residence, so only the wire
+    protocol differs from the production N>1 path.
+Combine-semantics are additionally covered by the gloo world-2 CPU
+tests (tests/test_dist_cpu.py), which share this code path."""
 
 import os
 import subprocess
