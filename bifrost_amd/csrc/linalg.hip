@@ -1825,6 +1825,215 @@ void cherk_ci8_mfma32_rs5_kernel(long n, long k, long nbatch, float alpha,
     }
 }
 
+/* ---- rs6: 8-wave 128x64-complex tile on v_mfma_i32_32x32x32_i8 -------- */
+// The issue-saturation analysis (profiles/round2_cherk.md closing
+// section) says non-MFMA instructions per CU-round are the remaining
+// cycle cost at occupancy 4.  rs6 keeps rs5's per-wave shape (32x32
+// complex, 8 mfma32/slab, 64 acc regs) but doubles the workgroup to 8
+// waves over a 128x64-complex tile: the I strip is shared by twice the
+// waves, cutting staged bytes per MFMA 25% (94 vs 125 B/equiv) and
+// halving barriers per MFMA.  Strides conflict-free as rs4/rs5
+// (I 288 B, J 160 B).  2 WGs/CU -> still 4 waves/SIMD (2 from each of
+// two independent workgroups).  Rectangular-triangular tile map (row
+// blocks 128, col blocks 64, J <= 2I+1).  Requirements as rs4.
+#define RS6_IROW 288
+#define RS6_JROW 160
+#define RS6_ISTRIP (64 * RS6_IROW)
+#define RS6_JSTRIP (64 * RS6_JROW)
+template <int SCHED>
+__global__ __launch_bounds__(512)
+__attribute__((amdgpu_waves_per_eu(4)))
+void cherk_ci8_mfma32_rs6_kernel(long n, long k, long nbatch, float alpha,
+                                 const signed char* __restrict__ a, long lda,
+                                 long a_b, float beta, f2* __restrict__ c,
+                                 long c_row, long c_b, long nti,
+                                 long ntiles) {
+    __shared__ signed char ldsI[2][64][RS6_IROW];
+    __shared__ signed char ldsJ[2][64][RS6_JROW];
+    int tid = threadIdx.x;
+    int lane = tid & 63;
+    int wave = tid >> 6;
+    int wr = wave >> 1, wc = wave & 1;  // 4x2 wave grid of 32x32c
+
+    int row32 = 16 * (lane >> 5) + ((lane >> 1) & 7);
+    int colb32 = 8 * ((lane & 1) + 2 * ((lane >> 4) & 1));
+
+    // staging: threads 0-127 stage the I strip (128 B each, the proven
+    // conflict-free 2-threads-per-row interleave), threads 128-255 the
+    // J strip (64 B each); threads 256-511 do not stage — their waves'
+    // MFMAs cover the stagers' issue on each SIMD.
+    bool st_on = tid < 256;
+    int st_isJ = (tid >> 7) & 1;
+    int tt = tid & 127;
+    int st_row = tt >> 1;
+    int st_h = tt & 1;
+
+    const signed char* rdI0 = &ldsI[0][0][0] + row32 * RS6_IROW + colb32
+                            + 64 * wr;
+    const signed char* rdJ0 = &ldsJ[0][0][0] + row32 * RS6_JROW + colb32
+                            + 64 * wc;
+    signed char* wrI0 = &ldsI[0][0][0] + st_row * RS6_IROW + 16 * st_h;
+    signed char* wrJ0 = &ldsJ[0][0][0] + st_row * RS6_JROW + 16 * st_h;
+
+    long total = 8 * ntiles * ((nbatch + 7) / 8);
+    for (long flat = blockIdx.x; flat < total; flat += gridDim.x) {
+        long q = flat >> 3, r8 = flat & 7;
+        long batch = r8 + 8 * (q / ntiles);
+        long t = q % ntiles;
+        if (batch >= nbatch) continue;
+        const signed char* ab = a + batch * a_b * 2;
+        f2* cb = c + batch * c_b;
+        long I = 0, rem = t;
+        while (rem >= (2 * I + 2 < nti ? 2 * I + 2 : nti)) {
+            rem -= (2 * I + 2 < nti ? 2 * I + 2 : nti);
+            ++I;
+        }
+        long J = rem;
+        long i0 = I * 128, j0 = J * 64;
+        long d = (j0 + 32 * wc) - (i0 + 32 * wr);
+        bool skip_all = d >= 32;           // quadrant fully above diag
+        bool diag_q = d == 0;              // on-diagonal quadrant
+        v16i acc[2][2];
+#pragma unroll
+        for (int x = 0; x < 2; ++x)
+#pragma unroll
+            for (int y = 0; y < 2; ++y) acc[x][y] = v16i{};
+
+        const long slab_step = 64 * lda * 2;
+        v4i stg[8];
+        const signed char* load_next = st_isJ
+            ? ab + (long)st_row * lda * 2 + j0 * 2 + 16 * st_h
+            : ab + (long)st_row * lda * 2 + i0 * 2 + 16 * st_h;
+        auto load_slab = [&]() {
+            if (!st_on) return;
+            const signed char* p = load_next;
+            load_next += slab_step;
+            if (st_isJ) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e)
+                    stg[e] = *(const v4i*)__builtin_assume_aligned(
+                        p + 32 * e, 16);
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    stg[e] = *(const v4i*)__builtin_assume_aligned(
+                        p + 32 * e, 16);
+            }
+        };
+        auto write_slab = [&](int buf) {
+            if (!st_on) return;
+            if (st_isJ) {
+                signed char* wb = wrJ0 + buf * RS6_JSTRIP;
+#pragma unroll
+                for (int e = 0; e < 4; ++e)
+                    *(v4i*)(wb + 16 * (2 * e)) = stg[e];
+            } else {
+                signed char* wb = wrI0 + buf * RS6_ISTRIP;
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    *(v4i*)(wb + 16 * (2 * e)) = stg[e];
+            }
+        };
+        auto fragI = [&](const signed char* base, int h, int cc) {
+            const signed char* p = base + (32 * h) * RS6_IROW + 32 * cc;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 8 * RS6_IROW));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto fragJ = [&](const signed char* base, int h, int cc) {
+            const signed char* p = base + (32 * h) * RS6_JROW + 32 * cc;
+            v2i lo = __builtin_amdgcn_ds_read_tr8_b64_v2i32((lds_v2i)p);
+            v2i hi = __builtin_amdgcn_ds_read_tr8_b64_v2i32(
+                (lds_v2i)(p + 8 * RS6_JROW));
+            return v4i{lo[0], lo[1], hi[0], hi[1]};
+        };
+        auto burst = [&](const signed char* bI, const signed char* bJ) {
+#pragma unroll
+            for (int h = 0; h < 2; ++h) {
+                v4i fb[2], fa[2];
+                fb[0] = fragJ(bJ, h, 0);
+                fb[1] = fragJ(bJ, h, 1);
+                fa[0] = fragI(bI, h, 0);
+                fa[1] = fragI(bI, h, 1);
+#pragma unroll
+                for (int ta = 0; ta < 2; ++ta)
+#pragma unroll
+                    for (int tb = 0; tb < 2; ++tb) {
+                        if (diag_q && ta < tb) continue;
+                        acc[ta][tb] = __builtin_amdgcn_mfma_i32_32x32x32_i8(
+                            fa[ta], fb[tb], acc[ta][tb], 0, 0, 0);
+                    }
+            }
+        };
+        auto step = [&](int buf, int s, int nslab) {
+            // the wave's quadrant base within the I strip: row half wr
+            const signed char* bI = rdI0 + buf * RS6_ISTRIP;
+            const signed char* bJ = rdJ0 + buf * RS6_JSTRIP;
+            if (SCHED == 0) {
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+                if (!skip_all) burst(bI, bJ);
+            } else {
+                if (!skip_all) burst(bI, bJ);
+                if (s + 1 < nslab) {
+                    write_slab(buf ^ 1);
+                    if (s + 2 < nslab) load_slab();
+                }
+            }
+            __syncthreads();
+        };
+
+        int nslab = (int)(k / 64);
+        load_slab();
+        write_slab(0);
+        if (nslab > 1) load_slab();
+        __syncthreads();
+        int s = 0;
+        while (s < nslab) {
+            step(0, s, nslab);
+            ++s;
+            if (s >= nslab) break;
+            step(1, s, nslab);
+            ++s;
+        }
+#pragma unroll
+        for (int ta = 0; ta < 2; ++ta) {
+#pragma unroll
+            for (int tb = 0; tb < 2; ++tb) {
+                if (skip_all || (diag_q && ta < tb)) continue;
+                long ci_base = i0 + 32 * wr + 16 * ta;
+                long cj = j0 + 32 * wc + 16 * tb + ((lane & 31) >> 1);
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+#pragma unroll
+                    for (int p = 0; p < 2; ++p) {
+                        int v0 = acc[ta][tb][4 * g + 2 * p];
+                        int v1 = acc[ta][tb][4 * g + 2 * p + 1];
+                        int sv0 = __shfl_xor(v0, 1);
+                        int sv1 = __shfl_xor(v1, 1);
+                        long i = ci_base + 4 * g + 2 * (lane >> 5) + p;
+                        long j = cj;
+                        bool write = (lane & 1) == 0 && i < n && j < n &&
+                                     i >= j;
+                        if (write) {
+                            float re = (float)(v0 + sv1);
+                            float im = (float)(sv0 - v1);
+                            f2 prev =
+                                beta != 0.f ? cb[i * c_row + j] : f2{};
+                            cb[i * c_row + j] =
+                                f2{alpha * re + beta * prev.x,
+                                   alpha * im + beta * prev.y};
+                        }
+                    }
+                }
+            }
+        }
+    }
+}
+
 /* -------- 8-wave rectangular-tile register-staged cherk (rs8) ----------- */
 // 512 threads / 8 waves per 128x64-complex output tile (wave grid 4x2,
 // each wave the same 32x32-complex quadrant as the rs kernel).  Staged
@@ -2591,12 +2800,41 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
             // the fallback for shapes rs5 cannot take and via
             // BIFROST_CHERK=rs.
             bool want_rs5 = !sel || strcmp(sel, "rs5") == 0;
+            bool want_rs6 = sel && strcmp(sel, "rs6") == 0 &&
+                            n % 128 == 0;
             bool want_rs2 = sel && strcmp(sel, "rs2") == 0 &&
                             n % 128 == 0;
             // rs8 (8-wave 128x64 tile) measures ~equal to rs (1.23 vs
             // 1.25 Gsamp/s at config 3) — opt-in until it wins.
             const char* sel8 = getenv("BIFROST_CHERK");
             bool want_rs8 = sel8 && strcmp(sel8, "rs8") == 0;
+            if (al16 && want_rs6) {
+                long nti6 = n / 64;
+                long nI6 = n / 128;
+                long ntiles6 = 0;
+                for (long I = 0; I < nI6; ++I)
+                    ntiles6 += (2 * I + 2 < nti6 ? 2 * I + 2 : nti6);
+                long nflat6 = ((ntiles6 * nbatch + 7) / 8) * 8;
+                dim3 grid6(cap_grid(nflat6, 65535), 1);
+                const char* schenv6 = getenv("BIFROST_CHERK_SCHED");
+                int sched6 = schenv6 ? atoi(schenv6) : 1;
+                if (sched6 == 0)
+                    hipLaunchKernelGGL(cherk_ci8_mfma32_rs6_kernel<0>,
+                                       grid6, dim3(512), 0, stream, n, k,
+                                       nbatch, (float)alpha,
+                                       (const signed char*)a, a_k, a_b,
+                                       (float)beta, (f2*)c, c_row, c_b,
+                                       nti6, ntiles6);
+                else
+                    hipLaunchKernelGGL(cherk_ci8_mfma32_rs6_kernel<1>,
+                                       grid6, dim3(512), 0, stream, n, k,
+                                       nbatch, (float)alpha,
+                                       (const signed char*)a, a_k, a_b,
+                                       (float)beta, (f2*)c, c_row, c_b,
+                                       nti6, ntiles6);
+                BF_CHECK_HIP(hipGetLastError());
+                return BF_STATUS_SUCCESS;
+            }
             if (al16 && want_rs5) {
                 const char* schenv5 = getenv("BIFROST_CHERK_SCHED");
                 // sched 1 (burst first, then writes+loads) wins the
