@@ -20,6 +20,7 @@ VARIANTS = [
     ("rs3", "0"), ("rs3", "1"),
     ("rs2", "0"), ("rs2", "1"),
     ("rs", "0"), ("rs", "2"), ("rs", "5"),
+    ("rs6", "0"), ("rs6", "1"),
     ("rs8", None), ("wave", None), ("coop", None), ("pipe", None),
 ]
 
