@@ -102,3 +102,41 @@ def test_c5_pipeline():
     assert got.shape == (1,) + gold.shape
     np.testing.assert_allclose(got.reshape(gold.shape), gold,
                                rtol=1e-3, atol=1e-2 * gold.max())
+
+
+def test_c5_bench_chain_device_accumulate():
+    """The bench.py --mode c5 configuration at small scale: all-device
+    rings, fft/detect gulping the whole spectra batch, accumulate as one
+    whole-window bfReduce on device (no host copy before the sink)."""
+    T, NC, NS, NB, NF = 128, 4, 32, 16, 32
+    n = NS * NPOL
+    rng = np.random.RandomState(77)
+    re = rng.randint(-7, 8, size=(T, NC, n))
+    im = rng.randint(-7, 8, size=(T, NC, n))
+    packed = (((re & 0xF) << 4) | (im & 0xF)).astype(np.uint8)
+    x = (re + 1j * im).astype(np.complex64)
+    voltages = bf.ndarray(packed.view(bf.DataType.ci4).reshape(T, NC, n))
+    w = (rng.standard_normal((NB, NC, n, 2)).astype(np.float32)
+         .view(np.complex64).reshape(NB, NC, n))
+
+    # gold: beamform -> fine FFT -> |.|^2 -> sum over spectra
+    y = np.einsum("bcn,tcn->tcb", w, x)
+    nspec = T // NF
+    spec = np.fft.fft(y.reshape(nspec, NF, NC, NB), axis=1)
+    gold = (np.abs(spec) ** 2).astype(np.float32).sum(axis=0)
+
+    out = []
+    with bf.Pipeline() as pipe:
+        src = NumpySourceBlock([voltages], gulp_nframe=T,
+                               labels=["time", "freq", "stand_pol"])
+        dev = bf.blocks.copy(src, space="cuda")
+        beam = BeamformBlock(dev, w)
+        fine = views.split_axis(beam, 0, NF, label="fine_time")
+        spec_b = bf.blocks.fft(fine, axes="fine_time", gulp_nframe=nspec)
+        pwr = bf.blocks.detect(spec_b, mode="scalar", gulp_nframe=nspec)
+        acc = bf.blocks.accumulate(pwr, nspec, gulp_nframe=nspec)
+        host = bf.blocks.copy(acc, space="system")
+        CollectBlock(host, out)
+        pipe.run()
+    got = np.concatenate(out, axis=0)[0]
+    np.testing.assert_allclose(got, gold, rtol=1e-3, atol=1e-2)
