@@ -2840,6 +2840,8 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 // sched 1 (burst first, then writes+loads) wins the
                 // same-box ABAB consistently (r2_ab9/r2_ab11)
                 int sched5 = schenv5 ? atoi(schenv5) : 1;
+                const char* genv5 = getenv("BIFROST_CHERK_GRID");
+                if (genv5) grid = dim3(cap_grid(atol(genv5), 65535), 1);
                 if (sched5 == 12)  // diagnostic: k-loop cycle dump
                     hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<12>, grid,
                                        dim3(256), 0, stream, n, k, nbatch,
