@@ -2772,7 +2772,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
             // L2 affinity; round up to a multiple of 8 so every XCD slot
             // participates in the remap.
             long nflat = ((ntiles * nbatch + 7) / 8) * 8;
-            dim3 grid(cap_grid(nflat, 65535), 1);
+            dim3 grid(cap_grid(nflat, 65528), 1);
             // Kernel selection: the wave-autonomous kernel (no workgroup
             // barriers, per-wave counted vmcnt) for aligned full-tile
             // shapes; the cooperative kernel for edges.  BIFROST_CHERK=coop
@@ -2815,7 +2815,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 for (long I = 0; I < nI6; ++I)
                     ntiles6 += (2 * I + 2 < nti6 ? 2 * I + 2 : nti6);
                 long nflat6 = ((ntiles6 * nbatch + 7) / 8) * 8;
-                dim3 grid6(cap_grid(nflat6, 65535), 1);
+                dim3 grid6(cap_grid(nflat6, 65528), 1);
                 const char* schenv6 = getenv("BIFROST_CHERK_SCHED");
                 int sched6 = schenv6 ? atoi(schenv6) : 1;
                 if (sched6 == 0)
@@ -2841,7 +2841,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 // same-box ABAB consistently (r2_ab9/r2_ab11)
                 int sched5 = schenv5 ? atoi(schenv5) : 1;
                 const char* genv5 = getenv("BIFROST_CHERK_GRID");
-                if (genv5) grid = dim3(cap_grid(atol(genv5), 65535), 1);
+                if (genv5) grid = dim3(cap_grid(atol(genv5), 65528), 1);
                 if (sched5 == 12)  // diagnostic: k-loop cycle dump
                     hipLaunchKernelGGL(cherk_ci8_mfma32_rs5_kernel<12>, grid,
                                        dim3(256), 0, stream, n, k, nbatch,
@@ -2876,9 +2876,9 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 for (long I = 0; I < nI4; ++I)
                     ntiles4 += (2 * I + 2 < nti4 ? 2 * I + 2 : nti4);
                 long nflat4 = ((ntiles4 * nbatch + 7) / 8) * 8;
-                dim3 grid4(cap_grid(nflat4, 65535), 1);
+                dim3 grid4(cap_grid(nflat4, 65528), 1);
                 const char* genv4 = getenv("BIFROST_CHERK_GRID");
-                if (genv4) grid4 = dim3(cap_grid(atol(genv4), 65535), 1);
+                if (genv4) grid4 = dim3(cap_grid(atol(genv4), 65528), 1);
                 const char* schenv4 = getenv("BIFROST_CHERK_SCHED");
                 int sched4 = schenv4 ? atoi(schenv4) : 0;
                 auto launch_rs4 = [&](auto kern) {
@@ -2899,11 +2899,11 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 long nb3 = n / 128;
                 long ntiles3 = nb3 * (nb3 + 1) / 2;
                 long nflat3 = ((ntiles3 * nbatch + 7) / 8) * 8;
-                dim3 grid3(cap_grid(nflat3, 65535), 1);
+                dim3 grid3(cap_grid(nflat3, 65528), 1);
                 const char* schenv3 = getenv("BIFROST_CHERK_SCHED");
                 int sched3 = schenv3 ? atoi(schenv3) : 1;
                 const char* genv = getenv("BIFROST_CHERK_GRID");
-                if (genv) grid3 = dim3(cap_grid(atol(genv), 65535), 1);
+                if (genv) grid3 = dim3(cap_grid(atol(genv), 65528), 1);
                 auto launch_rs3 = [&](auto kern) {
                     hipLaunchKernelGGL(kern, grid3, dim3(256), 0, stream, n,
                                        k, nbatch, (float)alpha,
@@ -2922,11 +2922,11 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 long nb2 = n / 128;
                 long ntiles2 = nb2 * (nb2 + 1) / 2;
                 long nflat2 = ((ntiles2 * nbatch + 7) / 8) * 8;
-                dim3 grid2(cap_grid(nflat2, 65535), 1);
+                dim3 grid2(cap_grid(nflat2, 65528), 1);
                 const char* schenv = getenv("BIFROST_CHERK_SCHED");
                 int sched = schenv ? atoi(schenv) : 0;
                 const char* genv2 = getenv("BIFROST_CHERK_GRID");
-                if (genv2) grid2 = dim3(cap_grid(atol(genv2), 65535), 1);
+                if (genv2) grid2 = dim3(cap_grid(atol(genv2), 65528), 1);
                 auto launch_rs2 = [&](auto kern) {
                     hipLaunchKernelGGL(kern, grid2, dim3(256), 0, stream, n,
                                        k, nbatch, (float)alpha,
@@ -2962,7 +2962,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 for (long I = 0; I < nI; ++I)
                     ntiles8 += (2 * I + 2 < nti ? 2 * I + 2 : nti);
                 long nflat8 = ((ntiles8 * nbatch + 7) / 8) * 8;
-                dim3 grid8(cap_grid(nflat8, 65535), 1);
+                dim3 grid8(cap_grid(nflat8, 65528), 1);
                 hipLaunchKernelGGL(cherk_ci8_mfma_rs8_kernel, grid8,
                                    dim3(512), 0, stream, n, k, nbatch,
                                    (float)alpha, (const signed char*)a, a_k,
@@ -3034,7 +3034,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
         }
         long ntiles_dim = (n + 31) / 32;
         long ntiles = ntiles_dim * (ntiles_dim + 1) / 2;
-        dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
+        dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65528));
         hipLaunchKernelGGL((cherk_ci8_kernel<32>), grid, dim3(256), 0, stream,
                            n, k, nbatch, (float)alpha,
                            (const signed char*)a, a_k, a_b, (float)beta,
@@ -3044,7 +3044,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
     }
     long ntiles_dim = (n + 15) / 16;
     long ntiles = ntiles_dim * (ntiles_dim + 1) / 2;
-    dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
+    dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65528));
     dim3 block(16, 16);
 #define HERK_CASE(LOADER, ACC)                                               \
     do {                                                                     \
@@ -3099,7 +3099,7 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
         (a_type == BF_DTYPE_CI16 || a_type == BF_DTYPE_CF32) &&
         c_type == BF_DTYPE_CF32 && a_k == 1 && b_k == 1 && !conj_a &&
         !conj_b && m <= 1024) {
-        dim3 grid(cap_grid((nn + 511) / 512, 4096), cap_grid(nbatch, 65535));
+        dim3 grid(cap_grid((nn + 511) / 512, 4096), cap_grid(nbatch, 65528));
         bool fast_ok = (k % 64 == 0) && (nn % 2 == 0) && nn >= 2;
         // bf16-split MFMA path (cf32 W, ci8 X, full tiles): ~1.25 PF
         // effective ceiling vs the f32 VALU kernels.  Opt out with
@@ -3110,7 +3110,7 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
             (a_type == BF_DTYPE_CF32 || a_type == BF_DTYPE_CI16) &&
             (b_type == BF_DTYPE_CI8 || b_type == BF_DTYPE_CI4) &&
             k % 64 == 0 && nn % 128 == 0 && m % 16 == 0 && nn > 0) {
-            dim3 mgrid(cap_grid(nn / 128, 4096), cap_grid(nbatch, 65535));
+            dim3 mgrid(cap_grid(nn / 128, 4096), cap_grid(nbatch, 65528));
             bool x4 = b_type == BF_DTYPE_CI4;
             bool w16 = a_type == BF_DTYPE_CI16;
             const char* ck = getenv("BIFROST_BEAM_CHUNK");
@@ -3129,7 +3129,7 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
 #define BEAM_MFMA_ONE(NBT, XTV, WTV)                                          \
     do {                                                                      \
         if (!(beam_sel && strcmp(beam_sel, "mfma2") == 0)) {                     \
-            dim3 mgrid3(cap_grid(nn / 64, 4096), cap_grid(nbatch, 65535));    \
+            dim3 mgrid3(cap_grid(nn / 64, 4096), cap_grid(nbatch, 65528));    \
             hipLaunchKernelGGL((beamform_mfma_kernel<NBT, XTV, WTV, 1, 0>),   \
                                mgrid3, dim3(256), 0, stream, nn, k, nbatch,   \
                                (float)alpha, a, a_i, a_b,                     \
@@ -3203,7 +3203,7 @@ BFstatus launch_gemm(BFdtype a_type, BFdtype b_type, BFdtype c_type, long m,
 
     long tiles_i = (m + 15) / 16, tiles_j = (nn + 15) / 16;
     long ntiles = tiles_i * tiles_j;
-    dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65535));
+    dim3 grid(cap_grid(ntiles, 16384), cap_grid(nbatch, 65528));
     dim3 block(16, 16);
 #define GEMM_CASE(LA, LB, ACC)                                                \
     do {                                                                      \
