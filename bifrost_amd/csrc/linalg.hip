@@ -2772,7 +2772,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
             // L2 affinity; round up to a multiple of 8 so every XCD slot
             // participates in the remap.
             long nflat = ((ntiles * nbatch + 7) / 8) * 8;
-            dim3 grid(cap_grid(nflat, 65528), 1);
+            dim3 grid(cap_grid(nflat, 1073741824), 1);
             // Kernel selection: the wave-autonomous kernel (no workgroup
             // barriers, per-wave counted vmcnt) for aligned full-tile
             // shapes; the cooperative kernel for edges.  BIFROST_CHERK=coop
@@ -2815,7 +2815,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 for (long I = 0; I < nI6; ++I)
                     ntiles6 += (2 * I + 2 < nti6 ? 2 * I + 2 : nti6);
                 long nflat6 = ((ntiles6 * nbatch + 7) / 8) * 8;
-                dim3 grid6(cap_grid(nflat6, 65528), 1);
+                dim3 grid6(cap_grid(nflat6, 1073741824), 1);
                 const char* schenv6 = getenv("BIFROST_CHERK_SCHED");
                 int sched6 = schenv6 ? atoi(schenv6) : 1;
                 if (sched6 == 0)
@@ -2876,7 +2876,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 for (long I = 0; I < nI4; ++I)
                     ntiles4 += (2 * I + 2 < nti4 ? 2 * I + 2 : nti4);
                 long nflat4 = ((ntiles4 * nbatch + 7) / 8) * 8;
-                dim3 grid4(cap_grid(nflat4, 65528), 1);
+                dim3 grid4(cap_grid(nflat4, 1073741824), 1);
                 const char* genv4 = getenv("BIFROST_CHERK_GRID");
                 if (genv4) grid4 = dim3(cap_grid(atol(genv4), 65528), 1);
                 const char* schenv4 = getenv("BIFROST_CHERK_SCHED");
@@ -2899,7 +2899,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 long nb3 = n / 128;
                 long ntiles3 = nb3 * (nb3 + 1) / 2;
                 long nflat3 = ((ntiles3 * nbatch + 7) / 8) * 8;
-                dim3 grid3(cap_grid(nflat3, 65528), 1);
+                dim3 grid3(cap_grid(nflat3, 1073741824), 1);
                 const char* schenv3 = getenv("BIFROST_CHERK_SCHED");
                 int sched3 = schenv3 ? atoi(schenv3) : 1;
                 const char* genv = getenv("BIFROST_CHERK_GRID");
@@ -2922,7 +2922,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 long nb2 = n / 128;
                 long ntiles2 = nb2 * (nb2 + 1) / 2;
                 long nflat2 = ((ntiles2 * nbatch + 7) / 8) * 8;
-                dim3 grid2(cap_grid(nflat2, 65528), 1);
+                dim3 grid2(cap_grid(nflat2, 1073741824), 1);
                 const char* schenv = getenv("BIFROST_CHERK_SCHED");
                 int sched = schenv ? atoi(schenv) : 0;
                 const char* genv2 = getenv("BIFROST_CHERK_GRID");
@@ -2962,7 +2962,7 @@ BFstatus launch_herk(BFdtype a_type, BFdtype c_type, long n, long k,
                 for (long I = 0; I < nI; ++I)
                     ntiles8 += (2 * I + 2 < nti ? 2 * I + 2 : nti);
                 long nflat8 = ((ntiles8 * nbatch + 7) / 8) * 8;
-                dim3 grid8(cap_grid(nflat8, 65528), 1);
+                dim3 grid8(cap_grid(nflat8, 1073741824), 1);
                 hipLaunchKernelGGL(cherk_ci8_mfma_rs8_kernel, grid8,
                                    dim3(512), 0, stream, n, k, nbatch,
                                    (float)alpha, (const signed char*)a, a_k,
